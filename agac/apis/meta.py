@@ -1,0 +1,138 @@
+"""Kubernetes-style object metadata and generic (de)serialization machinery.
+
+Replaces k8s.io/apimachinery for this framework: every API object is a
+dataclass with ``to_dict``/``from_dict`` producing the Kubernetes wire format
+(camelCase JSON), so objects round-trip through the HTTP API server and
+AdmissionReview payloads exactly like their Go counterparts.
+"""
+
+from __future__ import annotations
+
+import copy
+import dataclasses
+import typing
+from dataclasses import dataclass, field
+
+
+def _camelize(name: str) -> str:
+    parts = name.split("_")
+    return parts[0] + "".join(p[:1].upper() + p[1:] for p in parts[1:])
+
+
+def _is_api_type(tp) -> bool:
+    return dataclasses.is_dataclass(tp) and isinstance(tp, type)
+
+
+def _unwrap_optional(tp):
+    origin = typing.get_origin(tp)
+    if origin is typing.Union:
+        args = [a for a in typing.get_args(tp) if a is not type(None)]
+        if len(args) == 1:
+            return args[0]
+    return tp
+
+
+def _serialize(value):
+    if dataclasses.is_dataclass(value) and not isinstance(value, type):
+        return to_dict(value)
+    if isinstance(value, list):
+        return [_serialize(v) for v in value]
+    if isinstance(value, dict):
+        return {k: _serialize(v) for k, v in value.items()}
+    return value
+
+
+def to_dict(obj) -> dict:
+    """Serialize a dataclass API object to its camelCase wire format.
+
+    ``None`` fields are omitted (Go ``omitempty`` behavior); empty lists and
+    dicts are omitted too unless the field is listed in ``_keep_empty``.
+    """
+    overrides = getattr(type(obj), "_json_overrides", {})
+    keep_empty = getattr(type(obj), "_keep_empty", set())
+    out = {}
+    for f in dataclasses.fields(obj):
+        value = getattr(obj, f.name)
+        if value is None:
+            continue
+        if value in ({}, []) and f.name not in keep_empty:
+            continue
+        json_name = overrides.get(f.name, _camelize(f.name))
+        out[json_name] = _serialize(value)
+    return out
+
+
+def _deserialize(tp, value):
+    tp = _unwrap_optional(tp)
+    origin = typing.get_origin(tp)
+    if origin is list:
+        (item_tp,) = typing.get_args(tp)
+        return [_deserialize(item_tp, v) for v in value or []]
+    if origin is dict:
+        return dict(value or {})
+    if _is_api_type(tp):
+        return from_dict(tp, value or {})
+    return value
+
+
+def from_dict(cls, data: dict):
+    """Deserialize camelCase wire format into a dataclass API object.
+
+    Unknown keys are ignored (forward compatibility, like Go json decoding).
+    """
+    overrides = getattr(cls, "_json_overrides", {})
+    hints = typing.get_type_hints(cls)
+    kwargs = {}
+    for f in dataclasses.fields(cls):
+        json_name = overrides.get(f.name, _camelize(f.name))
+        if data is not None and json_name in data:
+            kwargs[f.name] = _deserialize(hints[f.name], data[json_name])
+    return cls(**kwargs)
+
+
+@dataclass
+class ObjectMeta:
+    """Reference: metav1.ObjectMeta (the subset the controllers use)."""
+
+    name: str = ""
+    namespace: str = ""
+    uid: str = ""
+    resource_version: str = ""
+    generation: int = 0
+    creation_timestamp: typing.Optional[str] = None
+    deletion_timestamp: typing.Optional[str] = None
+    annotations: typing.Dict[str, str] = field(default_factory=dict)
+    labels: typing.Dict[str, str] = field(default_factory=dict)
+    finalizers: typing.List[str] = field(default_factory=list)
+
+
+@dataclass
+class TypeMeta:
+    kind: str = ""
+    api_version: str = ""
+
+
+def deep_copy(obj):
+    """DeepCopy equivalent (reference uses generated DeepCopyObject)."""
+    return copy.deepcopy(obj)
+
+
+def meta_namespace_key(obj) -> str:
+    """cache.MetaNamespaceKeyFunc: '<namespace>/<name>' or '<name>'."""
+    meta = obj.metadata if hasattr(obj, "metadata") else obj
+    if meta.namespace:
+        return f"{meta.namespace}/{meta.name}"
+    return meta.name
+
+
+def split_meta_namespace_key(key: str):
+    """cache.SplitMetaNamespaceKey: returns (namespace, name).
+
+    Raises ValueError for keys with more than one '/'.
+    """
+    parts = key.split("/")
+    if len(parts) == 1:
+        return "", parts[0]
+    if len(parts) == 2:
+        return parts[0], parts[1]
+    raise ValueError(f"unexpected key format: {key!r}")

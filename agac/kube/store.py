@@ -1,0 +1,321 @@
+"""In-memory Kubernetes API server core: typed object store with optimistic
+concurrency, finalizer-aware deletion, generation tracking and watch streams.
+
+This is the stand-in for the kube-apiserver that the reference talks to via
+client-go.  All controller-visible semantics the reference relies on are
+modeled:
+
+- monotonically increasing ``resourceVersion`` per store, stamped on writes;
+  updates with a stale resourceVersion fail with ``ConflictError``
+  (needed for correct leader election and status-update races);
+- ``metadata.generation`` increments when ``spec`` changes (not on status
+  updates) — the EndpointGroupBinding reconciler gates on
+  ``observedGeneration == generation`` (reference ``egb/reconcile.go:157``);
+- deleting an object with finalizers only sets ``deletionTimestamp``; the
+  object is removed once the last finalizer is dropped via update
+  (reference relies on this for ``egb/reconcile.go:36-97``);
+- list+watch: ``list`` returns a consistent snapshot plus the store
+  resourceVersion, ``watch`` replays events after a given resourceVersion
+  from a bounded log, then streams live events (informer contract).
+"""
+
+from __future__ import annotations
+
+import itertools
+import queue
+import threading
+import time
+import uuid
+from collections import deque
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Tuple
+
+from ..apis import meta as metalib
+
+ADDED = "ADDED"
+MODIFIED = "MODIFIED"
+DELETED = "DELETED"
+
+# Bounded watch-event log; watches older than this are told to relist
+# (equivalent of apiserver's "410 Gone" / resourceVersion too old).
+_EVENT_LOG_SIZE = 4096
+
+
+class APIError(Exception):
+    def __init__(self, message: str, code: int):
+        super().__init__(message)
+        self.code = code
+
+
+class NotFoundError(APIError):
+    def __init__(self, message: str = "not found"):
+        super().__init__(message, 404)
+
+
+class AlreadyExistsError(APIError):
+    def __init__(self, message: str = "already exists"):
+        super().__init__(message, 409)
+
+
+class ConflictError(APIError):
+    def __init__(self, message: str = "resourceVersion conflict"):
+        super().__init__(message, 409)
+
+
+class GoneError(APIError):
+    """resourceVersion too old — caller must relist."""
+
+    def __init__(self, message: str = "resourceVersion too old"):
+        super().__init__(message, 410)
+
+
+def is_not_found(err: BaseException | None) -> bool:
+    return isinstance(err, NotFoundError)
+
+
+@dataclass
+class WatchEvent:
+    type: str
+    obj: object
+    resource_version: int
+
+
+class _Watch:
+    """A single watch subscription; iterated by informers."""
+
+    def __init__(self, store: "APIStore", kind: str, namespace: Optional[str]):
+        self._queue: "queue.Queue[Optional[WatchEvent]]" = queue.Queue()
+        self._store = store
+        self._kind = kind
+        self._namespace = namespace
+        self._stopped = False
+
+    def _accepts(self, kind: str, namespace: str) -> bool:
+        if kind != self._kind:
+            return False
+        return self._namespace is None or namespace == self._namespace
+
+    def _push(self, event: WatchEvent):
+        self._queue.put(event)
+
+    def stop(self):
+        if not self._stopped:
+            self._stopped = True
+            self._store._remove_watch(self)
+            self._queue.put(None)
+
+    def __iter__(self):
+        return self
+
+    def __next__(self) -> WatchEvent:
+        item = self._queue.get()
+        if item is None:
+            raise StopIteration
+        return item
+
+    def get(self, timeout: Optional[float] = None) -> Optional[WatchEvent]:
+        try:
+            return self._queue.get(timeout=timeout)
+        except queue.Empty:
+            return None
+
+
+class APIStore:
+    """Thread-safe multi-kind object store with watch semantics."""
+
+    def __init__(self):
+        self._lock = threading.RLock()
+        self._rv = itertools.count(1)
+        # kind -> {(namespace, name) -> obj}
+        self._objects: Dict[str, Dict[Tuple[str, str], object]] = {}
+        self._watches: List[_Watch] = []
+        self._event_log: deque = deque(maxlen=_EVENT_LOG_SIZE)
+
+    # -- helpers -----------------------------------------------------------
+    def _bucket(self, kind: str) -> Dict[Tuple[str, str], object]:
+        return self._objects.setdefault(kind, {})
+
+    def _next_rv(self) -> int:
+        return next(self._rv)
+
+    def _broadcast(self, kind: str, event_type: str, obj, rv: int):
+        event = WatchEvent(event_type, obj, rv)
+        self._event_log.append((kind, event))
+        for w in list(self._watches):
+            if w._accepts(kind, obj.metadata.namespace):
+                w._push(WatchEvent(event_type, metalib.deep_copy(obj), rv))
+
+    def _remove_watch(self, w: _Watch):
+        with self._lock:
+            if w in self._watches:
+                self._watches.remove(w)
+
+    # -- CRUD --------------------------------------------------------------
+    def create(self, obj):
+        kind = type(obj).kind
+        with self._lock:
+            key = (obj.metadata.namespace, obj.metadata.name)
+            bucket = self._bucket(kind)
+            if key in bucket:
+                raise AlreadyExistsError(f"{kind} {key} already exists")
+            stored = metalib.deep_copy(obj)
+            rv = self._next_rv()
+            stored.metadata.resource_version = str(rv)
+            stored.metadata.uid = stored.metadata.uid or str(uuid.uuid4())
+            stored.metadata.creation_timestamp = (
+                stored.metadata.creation_timestamp
+                or time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+            )
+            if hasattr(stored, "spec"):
+                stored.metadata.generation = 1
+            bucket[key] = stored
+            self._broadcast(kind, ADDED, stored, rv)
+            return metalib.deep_copy(stored)
+
+    def get(self, kind: str, namespace: str, name: str):
+        with self._lock:
+            obj = self._bucket(kind).get((namespace, name))
+            if obj is None:
+                raise NotFoundError(f"{kind} {namespace}/{name} not found")
+            return metalib.deep_copy(obj)
+
+    def list(self, kind: str, namespace: Optional[str] = None):
+        """Returns (objects, resourceVersion) — a consistent snapshot."""
+        with self._lock:
+            items = [
+                metalib.deep_copy(o)
+                for (ns, _), o in sorted(self._bucket(kind).items())
+                if namespace is None or ns == namespace
+            ]
+            # store-level rv: highest issued so far
+            rv = next(self._rv)  # burn one to get a strictly usable marker
+            return items, rv
+
+    def _check_rv(self, existing, obj):
+        if (
+            obj.metadata.resource_version
+            and obj.metadata.resource_version != existing.metadata.resource_version
+        ):
+            raise ConflictError(
+                f"{type(obj).kind} {obj.metadata.namespace}/{obj.metadata.name}: "
+                f"resourceVersion {obj.metadata.resource_version} != "
+                f"{existing.metadata.resource_version}"
+            )
+
+    def update(self, obj):
+        """Update spec+metadata.  Bumps generation if spec changed; removes the
+        object if it has a deletionTimestamp and finalizers became empty."""
+        kind = type(obj).kind
+        with self._lock:
+            key = (obj.metadata.namespace, obj.metadata.name)
+            bucket = self._bucket(kind)
+            existing = bucket.get(key)
+            if existing is None:
+                raise NotFoundError(f"{kind} {key} not found")
+            self._check_rv(existing, obj)
+
+            stored = metalib.deep_copy(obj)
+            # immutable/system-owned fields
+            stored.metadata.uid = existing.metadata.uid
+            stored.metadata.creation_timestamp = existing.metadata.creation_timestamp
+            stored.metadata.deletion_timestamp = existing.metadata.deletion_timestamp
+            stored.metadata.generation = existing.metadata.generation
+            if hasattr(stored, "status"):
+                stored.status = metalib.deep_copy(existing.status)
+            if hasattr(stored, "spec") and metalib.to_dict(stored.spec) != metalib.to_dict(
+                existing.spec
+            ):
+                stored.metadata.generation += 1
+
+            if (
+                stored.metadata.deletion_timestamp is not None
+                and not stored.metadata.finalizers
+            ):
+                rv = self._next_rv()
+                stored.metadata.resource_version = str(rv)
+                del bucket[key]
+                self._broadcast(kind, DELETED, stored, rv)
+                return metalib.deep_copy(stored)
+
+            rv = self._next_rv()
+            stored.metadata.resource_version = str(rv)
+            bucket[key] = stored
+            self._broadcast(kind, MODIFIED, stored, rv)
+            return metalib.deep_copy(stored)
+
+    def update_status(self, obj):
+        """Status-subresource update: only ``status`` is taken from ``obj``."""
+        kind = type(obj).kind
+        with self._lock:
+            key = (obj.metadata.namespace, obj.metadata.name)
+            bucket = self._bucket(kind)
+            existing = bucket.get(key)
+            if existing is None:
+                raise NotFoundError(f"{kind} {key} not found")
+            self._check_rv(existing, obj)
+            stored = metalib.deep_copy(existing)
+            stored.status = metalib.deep_copy(obj.status)
+            rv = self._next_rv()
+            stored.metadata.resource_version = str(rv)
+            bucket[key] = stored
+            self._broadcast(kind, MODIFIED, stored, rv)
+            return metalib.deep_copy(stored)
+
+    def delete(self, kind: str, namespace: str, name: str):
+        """Finalizer-aware delete (kube-apiserver graceful deletion)."""
+        with self._lock:
+            key = (namespace, name)
+            bucket = self._bucket(kind)
+            existing = bucket.get(key)
+            if existing is None:
+                raise NotFoundError(f"{kind} {namespace}/{name} not found")
+            if existing.metadata.finalizers:
+                if existing.metadata.deletion_timestamp is None:
+                    stored = metalib.deep_copy(existing)
+                    stored.metadata.deletion_timestamp = time.strftime(
+                        "%Y-%m-%dT%H:%M:%SZ", time.gmtime()
+                    )
+                    rv = self._next_rv()
+                    stored.metadata.resource_version = str(rv)
+                    bucket[key] = stored
+                    self._broadcast(kind, MODIFIED, stored, rv)
+                return None
+            rv = self._next_rv()
+            removed = metalib.deep_copy(existing)
+            removed.metadata.resource_version = str(rv)
+            del bucket[key]
+            self._broadcast(kind, DELETED, removed, rv)
+            return None
+
+    # -- watch -------------------------------------------------------------
+    def watch(
+        self,
+        kind: str,
+        namespace: Optional[str] = None,
+        resource_version: Optional[int] = None,
+    ) -> _Watch:
+        """Subscribe to events for ``kind``.  If ``resource_version`` is given,
+        replays logged events with rv > resource_version first; raises
+        ``GoneError`` if that window has been evicted from the log."""
+        with self._lock:
+            w = _Watch(self, kind, namespace)
+            if resource_version is not None:
+                logged = [
+                    (k, e) for (k, e) in self._event_log if k == kind
+                ]
+                if logged and logged[0][1].resource_version > resource_version + 1:
+                    # The requested window may predate the log.  Only an issue
+                    # if events were actually evicted; be conservative.
+                    if len(self._event_log) == self._event_log.maxlen:
+                        raise GoneError()
+                for _, e in logged:
+                    if e.resource_version > resource_version and (
+                        namespace is None or e.obj.metadata.namespace == namespace
+                    ):
+                        w._push(
+                            WatchEvent(
+                                e.type, metalib.deep_copy(e.obj), e.resource_version
+                            )
+                        )
+            self._watches.append(w)
+            return w

@@ -1,0 +1,50 @@
+"""Prometheus metrics.
+
+The reference has no metrics endpoint (SURVEY.md §5 flags this as a gap);
+this build closes it: reconcile outcomes/latency, workqueue depth and AWS
+API call counters, exposed via ``start_metrics_server``.  All metric calls
+are no-ops if prometheus_client is unavailable.
+"""
+
+from __future__ import annotations
+
+try:
+    from prometheus_client import Counter, Histogram, start_http_server
+
+    _AVAILABLE = True
+except Exception:  # pragma: no cover - prometheus_client is installed here
+    _AVAILABLE = False
+
+if _AVAILABLE:
+    RECONCILE_TOTAL = Counter(
+        "agac_reconcile_total",
+        "Reconcile attempts by queue and outcome",
+        ["queue", "outcome"],
+    )
+    RECONCILE_DURATION = Histogram(
+        "agac_reconcile_duration_seconds",
+        "Reconcile latency by queue",
+        ["queue"],
+        buckets=(0.0005, 0.001, 0.005, 0.01, 0.05, 0.1, 0.5, 1.0, 5.0, 30.0),
+    )
+    AWS_API_CALLS = Counter(
+        "agac_aws_api_calls_total",
+        "AWS API operations issued",
+        ["service", "operation"],
+    )
+
+
+def observe_reconcile(queue_name: str, outcome: str, seconds: float):
+    if _AVAILABLE:
+        RECONCILE_TOTAL.labels(queue=queue_name or "unknown", outcome=outcome).inc()
+        RECONCILE_DURATION.labels(queue=queue_name or "unknown").observe(seconds)
+
+
+def observe_aws_call(service: str, operation: str):
+    if _AVAILABLE:
+        AWS_API_CALLS.labels(service=service, operation=operation).inc()
+
+
+def start_metrics_server(port: int):
+    if _AVAILABLE:
+        start_http_server(port)

@@ -1,0 +1,134 @@
+"""Reconcile engine tests (reference pkg/reconcile/reconcile.go semantics).
+
+The reference has no unit tests for this engine (SURVEY.md §4 gap); these
+tests pin down every branch of the Result/error handling table.
+"""
+
+from agac.errors import NoRetryError
+from agac.kube.store import NotFoundError
+from agac.kube.workqueue import ItemExponentialFailureRateLimiter, RateLimitingQueue
+from agac.reconcile import Result, process_next_work_item
+
+
+class Obj:
+    def __init__(self, name):
+        self.name = name
+
+    def __deepcopy__(self, memo):  # engine deep-copies before processing
+        return Obj(self.name)
+
+
+def make_queue():
+    return RateLimitingQueue(
+        rate_limiter=ItemExponentialFailureRateLimiter(0.001, 0.01), name="t"
+    )
+
+
+def pump(q, key_to_obj, on_delete, on_update):
+    q.add("default/x")
+    return process_next_work_item(q, key_to_obj, on_delete, on_update)
+
+
+def test_create_or_update_called_with_copy():
+    q = make_queue()
+    calls = []
+    original = Obj("x")
+    pump(q, lambda k: original, lambda k: Result(), lambda o: calls.append(o) or Result())
+    assert len(calls) == 1
+    assert calls[0] is not original  # DeepCopyObject before processing
+
+
+def test_not_found_routes_to_delete():
+    q = make_queue()
+    calls = []
+
+    def key_to_obj(key):
+        raise NotFoundError()
+
+    pump(q, key_to_obj, lambda k: calls.append(k) or Result(), lambda o: Result())
+    assert calls == ["default/x"]
+
+
+def test_lookup_error_is_not_delete():
+    q = make_queue()
+    deleted = []
+
+    def key_to_obj(key):
+        raise RuntimeError("store broken")
+
+    assert pump(q, key_to_obj, lambda k: deleted.append(k) or Result(), lambda o: Result())
+    assert deleted == []
+
+
+def test_error_requeues_rate_limited():
+    q = make_queue()
+
+    def boom(obj):
+        raise RuntimeError("transient")
+
+    pump(q, lambda k: Obj("x"), lambda k: Result(), boom)
+    q.done("default/x")
+    item, shutdown = q.get(timeout=2.0)
+    assert item == "default/x" and not shutdown
+    assert q.num_requeues("default/x") >= 1
+
+
+def test_no_retry_error_forgets():
+    q = make_queue()
+
+    def boom(obj):
+        raise NoRetryError("permanent")
+
+    pump(q, lambda k: Obj("x"), lambda k: Result(), boom)
+    q.done("default/x")
+    item, _ = q.get(timeout=0.05)
+    assert item is None  # not requeued
+
+
+def test_requeue_after_uses_add_after_and_forgets():
+    q = make_queue()
+    results = iter([Result(requeue=True, requeue_after=0.02), Result()])
+    seen = []
+
+    def process(obj):
+        seen.append(obj.name)
+        return next(results)
+
+    pump(q, lambda k: Obj("x"), lambda k: Result(), process)
+    q.done("default/x")
+    assert q.num_requeues("default/x") == 0  # forgotten before AddAfter
+    item, _ = q.get(timeout=2.0)
+    assert item == "default/x"
+
+
+def test_requeue_flag_rate_limits():
+    q = make_queue()
+    pump(q, lambda k: Obj("x"), lambda k: Result(), lambda o: Result(requeue=True))
+    q.done("default/x")
+    item, _ = q.get(timeout=2.0)
+    assert item == "default/x"
+    assert q.num_requeues("default/x") >= 1
+
+
+def test_success_forgets():
+    q = make_queue()
+    pump(q, lambda k: Obj("x"), lambda k: Result(), lambda o: Result())
+    q.done("default/x")
+    assert q.num_requeues("default/x") == 0
+    item, _ = q.get(timeout=0.05)
+    assert item is None
+
+
+def test_shutdown_stops_worker():
+    q = make_queue()
+    q.shut_down()
+    assert not process_next_work_item(q, lambda k: None, lambda k: Result(), lambda o: Result())
+
+
+def test_non_string_key_is_forgotten():
+    q = make_queue()
+    q.add(42)
+    assert process_next_work_item(q, lambda k: None, lambda k: Result(), lambda o: Result())
+    q.done(42)
+    item, _ = q.get(timeout=0.05)
+    assert item is None
