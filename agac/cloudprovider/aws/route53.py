@@ -1,0 +1,275 @@
+"""Route53 resource manager.
+
+Behavior parity with reference ``pkg/cloudprovider/aws/route53.go``:
+A-ALIAS records pointing at the Global Accelerator DNS name, paired with
+TXT ownership records whose value identifies the managing cluster and the
+owning k8s resource.  Hosted zones are found by walking up parent domains;
+wildcard names round-trip through Route53's ``\\052`` octal escaping.
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import List, Optional, Tuple
+
+from ... import metrics
+from . import types as t
+
+logger = logging.getLogger(__name__)
+
+GA_MISSING_RETRY = 60.0  # seconds (reference route53.go:72-76)
+TXT_TTL = 300  # seconds (reference :276)
+
+
+# ---------------------------------------------------------------------------
+# Pure helpers (covered by the reference's unit table route53_test.go)
+# ---------------------------------------------------------------------------
+def route53_owner_value(cluster_name: str, resource: str, ns: str, name: str) -> str:
+    """TXT ownership record value — byte-for-byte parity including the
+    embedded quotes (reference ``Route53OwnerValue``, route53.go:18-20)."""
+    return (
+        '"heritage=aws-global-accelerator-controller,cluster='
+        + cluster_name
+        + ","
+        + resource
+        + "/"
+        + ns
+        + "/"
+        + name
+        + '"'
+    )
+
+
+def replace_wildcards(s: str) -> str:
+    """Route53 returns '*' as octal '\\052'; undo the first occurrence."""
+    return s.replace("\\052", "*", 1)
+
+
+def parent_domain(hostname: str) -> str:
+    """Strip the leftmost label ('a.b.c' → 'b.c', 'c' → '')."""
+    return ".".join(hostname.split(".")[1:])
+
+
+def find_a_record(
+    records: List[t.ResourceRecordSet], hostname: str
+) -> Optional[t.ResourceRecordSet]:
+    for record in records:
+        if record.type == t.RR_TYPE_A and replace_wildcards(record.name) == hostname + ".":
+            return record
+    return None
+
+
+def need_records_update(record: t.ResourceRecordSet, accelerator) -> bool:
+    """True when the A-alias target drifted from the accelerator DNS name
+    (reference :373-381; the stored alias DNS name is dot-terminated)."""
+    if record.alias_target is None:
+        return True
+    return record.alias_target.dns_name != accelerator.dns_name + "."
+
+
+# ---------------------------------------------------------------------------
+# Resource manager
+# ---------------------------------------------------------------------------
+class Route53Mixin:
+    """Methods bound into ``agac.cloudprovider.aws.client.AWS``."""
+
+    def ensure_route53_for_service(
+        self, svc, lb_ingress, hostnames: List[str], cluster_name: str
+    ) -> Tuple[bool, float]:
+        return self._ensure_route53(
+            lb_ingress.hostname,
+            hostnames,
+            cluster_name,
+            "service",
+            svc.metadata.namespace,
+            svc.metadata.name,
+        )
+
+    def ensure_route53_for_ingress(
+        self, ingress, lb_ingress, hostnames: List[str], cluster_name: str
+    ) -> Tuple[bool, float]:
+        return self._ensure_route53(
+            lb_ingress.hostname,
+            hostnames,
+            cluster_name,
+            "ingress",
+            ingress.metadata.namespace,
+            ingress.metadata.name,
+        )
+
+    def _ensure_route53(
+        self,
+        lb_hostname: str,
+        hostnames: List[str],
+        cluster_name: str,
+        resource: str,
+        ns: str,
+        name: str,
+    ) -> Tuple[bool, float]:
+        """Returns (created, retry_after_seconds).  0 or >1 matching
+        accelerators ⇒ requeue after 60s (reference :62-78)."""
+        accelerators = self.list_global_accelerator_by_hostname(
+            lb_hostname, cluster_name
+        )
+        if len(accelerators) > 1:
+            logger.error("Too many Global Accelerators for %s", lb_hostname)
+            return False, GA_MISSING_RETRY
+        if not accelerators:
+            logger.error("Could not find Global Accelerator for %s", lb_hostname)
+            return False, GA_MISSING_RETRY
+        accelerator = accelerators[0]
+
+        owner_value = route53_owner_value(cluster_name, resource, ns, name)
+        created = False
+        for hostname in hostnames:
+            hosted_zone = self.get_hosted_zone(hostname)
+            logger.info("HostedZone is %s", hosted_zone.id)
+            records = self.find_owned_a_record_sets(hosted_zone, owner_value)
+            record = find_a_record(records, hostname)
+            if record is None:
+                logger.info(
+                    "Creating record for %s with %s",
+                    hostname,
+                    accelerator.accelerator_arn,
+                )
+                self._create_metadata_record_set(
+                    hosted_zone, hostname, owner_value
+                )
+                self._create_record_set(hosted_zone, hostname, accelerator)
+                created = True
+            else:
+                if not need_records_update(record, accelerator):
+                    logger.info("Do not need to update for %s, so skip it", record.name)
+                    continue
+                self._update_record_set(hosted_zone, hostname, accelerator)
+                logger.info("RecordSet %s is updated", record.name)
+
+        logger.info("All records are synced for %s %s/%s", resource, ns, name)
+        return created, 0.0
+
+    def cleanup_record_set(
+        self, cluster_name: str, resource: str, ns: str, name: str
+    ):
+        """Scan every zone; delete owned A-alias records then their TXT
+        ownership records (reference :132-165)."""
+        owner_value = route53_owner_value(cluster_name, resource, ns, name)
+        for zone in self._list_all_hosted_zones():
+            for record in self.find_owned_a_record_sets(zone, owner_value):
+                self._delete_record(zone, record)
+                logger.info("Record set %s: %s is deleted", record.name, record.type)
+            for record in self._find_owned_metadata_record_sets(zone, owner_value):
+                self._delete_record(zone, record)
+                logger.info("Record set %s: %s is deleted", record.name, record.type)
+
+    # -- record discovery ---------------------------------------------------
+    def find_owned_a_record_sets(
+        self, hosted_zone: t.HostedZone, owner_value: str
+    ) -> List[t.ResourceRecordSet]:
+        """Alias record sets whose name has a TXT record carrying our
+        ownership value (reference ``FindOwneredARecordSets``, :216-238)."""
+        record_sets = self._list_record_sets(hosted_zone.id)
+        owned_names = [
+            rs.name
+            for rs in record_sets
+            for record in rs.resource_records
+            if record.value == owner_value
+        ]
+        return [
+            rs
+            for rs in record_sets
+            if rs.name in owned_names and rs.alias_target is not None
+        ]
+
+    def _find_owned_metadata_record_sets(
+        self, hosted_zone: t.HostedZone, owner_value: str
+    ) -> List[t.ResourceRecordSet]:
+        return [
+            rs
+            for rs in self._list_record_sets(hosted_zone.id)
+            for record in rs.resource_records
+            if record.value == owner_value
+        ]
+
+    # -- zone discovery ------------------------------------------------------
+    def get_hosted_zone(self, original_hostname: str) -> t.HostedZone:
+        """Walk up parent domains until a hosted zone matches
+        (reference ``GetHostedZone``, :335-358)."""
+        target = original_hostname
+        while True:
+            if not target:
+                raise ValueError(
+                    f"Could not find hosted zone for {original_hostname}"
+                )
+            logger.debug("Getting hosted zone for %s", target)
+            metrics.observe_aws_call("route53", "ListHostedZonesByName")
+            zones = self.route53.list_hosted_zones_by_name(
+                dns_name=target + ".", max_items=1
+            )
+            for zone in zones:
+                if zone.name == target + ".":
+                    return zone
+            target = parent_domain(target)
+
+    def _list_all_hosted_zones(self) -> List[t.HostedZone]:
+        zones: List[t.HostedZone] = []
+        token = None
+        while True:
+            metrics.observe_aws_call("route53", "ListHostedZones")
+            page, token = self.route53.list_hosted_zones(max_items=100, marker=token)
+            zones.extend(page)
+            if token is None:
+                return zones
+
+    def _list_record_sets(self, zone_id: str) -> List[t.ResourceRecordSet]:
+        records: List[t.ResourceRecordSet] = []
+        token = None
+        while True:
+            metrics.observe_aws_call("route53", "ListResourceRecordSets")
+            page, token = self.route53.list_resource_record_sets(
+                zone_id, max_items=300, start_token=token
+            )
+            records.extend(page)
+            if token is None:
+                return records
+
+    # -- record mutations ----------------------------------------------------
+    def _change(self, zone: t.HostedZone, action: str, record_set: t.ResourceRecordSet):
+        metrics.observe_aws_call("route53", "ChangeResourceRecordSets")
+        self.route53.change_resource_record_sets(
+            zone.id, [t.Change(action=action, record_set=record_set)]
+        )
+
+    def _alias_record_set(self, hostname: str, accelerator) -> t.ResourceRecordSet:
+        return t.ResourceRecordSet(
+            name=hostname,
+            type=t.RR_TYPE_A,
+            alias_target=t.AliasTarget(
+                dns_name=accelerator.dns_name,
+                evaluate_target_health=True,
+                # every Global Accelerator lives in this fixed alias zone
+                hosted_zone_id=t.GLOBAL_ACCELERATOR_HOSTED_ZONE_ID,
+            ),
+        )
+
+    def _create_record_set(self, zone: t.HostedZone, hostname: str, accelerator):
+        self._change(zone, t.CHANGE_ACTION_CREATE, self._alias_record_set(hostname, accelerator))
+
+    def _update_record_set(self, zone: t.HostedZone, hostname: str, accelerator):
+        self._change(zone, t.CHANGE_ACTION_UPSERT, self._alias_record_set(hostname, accelerator))
+
+    def _create_metadata_record_set(
+        self, zone: t.HostedZone, hostname: str, owner_value: str
+    ):
+        self._change(
+            zone,
+            t.CHANGE_ACTION_CREATE,
+            t.ResourceRecordSet(
+                name=hostname,
+                type=t.RR_TYPE_TXT,
+                ttl=TXT_TTL,
+                resource_records=[t.ResourceRecord(value=owner_value)],
+            ),
+        )
+
+    def _delete_record(self, zone: t.HostedZone, record: t.ResourceRecordSet):
+        self._change(zone, t.CHANGE_ACTION_DELETE, record)

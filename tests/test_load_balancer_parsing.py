@@ -1,0 +1,91 @@
+"""Hostname → (name, region) parsing parity
+(reference pkg/cloudprovider/aws/load_balancer_test.go)."""
+
+import pytest
+
+from agac.cloudprovider import detect_cloud_provider
+from agac.cloudprovider.aws import get_lb_name_from_hostname, get_region_from_arn
+
+
+class TestALBHostnames:
+    def test_public_alb(self):
+        name, region = get_lb_name_from_hostname(
+            "myapp-ingress-1234567890.ap-northeast-1.elb.amazonaws.com"
+        )
+        assert name == "myapp-ingress"
+        assert region == "ap-northeast-1"
+
+    def test_internal_alb(self):
+        name, region = get_lb_name_from_hostname(
+            "internal-myapp-ingress-1234567890.us-east-1.elb.amazonaws.com"
+        )
+        assert name == "myapp-ingress"
+        assert region == "us-east-1"
+
+    def test_alb_single_word_name(self):
+        name, region = get_lb_name_from_hostname(
+            "web-abc123.eu-west-1.elb.amazonaws.com"
+        )
+        assert name == "web"
+        assert region == "eu-west-1"
+
+
+class TestNLBHostnames:
+    def test_nlb(self):
+        name, region = get_lb_name_from_hostname(
+            "myservice-nlb-0123456789abcdef.elb.ap-northeast-1.amazonaws.com"
+        )
+        assert name == "myservice-nlb"
+        assert region == "ap-northeast-1"
+
+    def test_nlb_simple(self):
+        name, region = get_lb_name_from_hostname(
+            "api-aabbccdd00112233.elb.us-west-2.amazonaws.com"
+        )
+        assert name == "api"
+        assert region == "us-west-2"
+
+
+class TestNonELB:
+    @pytest.mark.parametrize(
+        "hostname",
+        [
+            "example.com",
+            "foo.cloudfront.net",
+            "s3.amazonaws.com",
+            "myapp.us-east-1.rds.amazonaws.com",
+        ],
+    )
+    def test_rejected(self, hostname):
+        with pytest.raises(ValueError):
+            get_lb_name_from_hostname(hostname)
+
+    def test_unparseable_subdomain(self):
+        # no '-hash' suffix → parse failure
+        with pytest.raises(ValueError):
+            get_lb_name_from_hostname("justone.us-east-1.elb.amazonaws.com")
+
+
+def test_get_region_from_arn():
+    assert (
+        get_region_from_arn(
+            "arn:aws:elasticloadbalancing:ap-northeast-1:123456789012:loadbalancer/net/x/abc"
+        )
+        == "ap-northeast-1"
+    )
+    assert get_region_from_arn("arn:aws:globalaccelerator::123:accelerator/a") == ""
+
+
+class TestDetectCloudProvider:
+    def test_aws(self):
+        assert (
+            detect_cloud_provider("foo-123.elb.us-east-1.amazonaws.com") == "aws"
+        )
+
+    def test_unknown(self):
+        with pytest.raises(ValueError):
+            detect_cloud_provider("foo.example.org")
+
+    def test_short_hostname(self):
+        with pytest.raises(ValueError):
+            detect_cloud_provider("localhost")
