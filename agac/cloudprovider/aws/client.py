@@ -42,6 +42,8 @@ class AWS(LoadBalancerMixin, GlobalAcceleratorMixin, Route53Mixin):
         poll_interval: float = 10.0,
         poll_timeout: float = 180.0,
         sleep: Callable[[float], None] = time.sleep,
+        lb_not_active_retry: float = 30.0,
+        ga_missing_retry: float = 60.0,
     ):
         self.lb = lb
         self.ga = ga
@@ -50,6 +52,10 @@ class AWS(LoadBalancerMixin, GlobalAcceleratorMixin, Route53Mixin):
         self.poll_interval = poll_interval
         self.poll_timeout = poll_timeout
         self.sleep = sleep
+        # requeue intervals (reference hardcodes 30s / 60s; injectable here
+        # so hermetic tests exercise the retry paths without wall-clock waits)
+        self.lb_not_active_retry = lb_not_active_retry
+        self.ga_missing_retry = ga_missing_retry
 
 
 # factory(region) -> AWS
@@ -65,6 +71,8 @@ class FakeCloudFactory:
         poll_interval: float = 0.0,
         poll_timeout: float = 5.0,
         sleep: Optional[Callable[[float], None]] = None,
+        lb_not_active_retry: float = 30.0,
+        ga_missing_retry: float = 60.0,
     ):
         if backend is None:
             from ..fake import FakeAWSBackend
@@ -74,6 +82,8 @@ class FakeCloudFactory:
         self.poll_interval = poll_interval
         self.poll_timeout = poll_timeout
         self.sleep = sleep if sleep is not None else (lambda s: None)
+        self.lb_not_active_retry = lb_not_active_retry
+        self.ga_missing_retry = ga_missing_retry
 
     def __call__(self, region: str) -> AWS:
         return AWS(
@@ -84,6 +94,8 @@ class FakeCloudFactory:
             poll_interval=self.poll_interval,
             poll_timeout=self.poll_timeout,
             sleep=self.sleep,
+            lb_not_active_retry=self.lb_not_active_retry,
+            ga_missing_retry=self.ga_missing_retry,
         )
 
 

@@ -40,7 +40,7 @@ GLOBAL_ACCELERATOR_OWNER_TAG_KEY = "aws-global-accelerator-owner"
 GLOBAL_ACCELERATOR_TARGET_HOSTNAME_KEY = "aws-global-accelerator-target-hostname"
 GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY = "aws-global-accelerator-cluster"
 
-LB_NOT_ACTIVE_RETRY = 30.0  # seconds (reference :127)
+DEFAULT_LB_NOT_ACTIVE_RETRY = 30.0  # seconds (reference :127)
 
 
 # ---------------------------------------------------------------------------
@@ -269,7 +269,7 @@ class GlobalAcceleratorMixin:
             logger.warning(
                 "LoadBalancer %s is not Active: %s", lb.load_balancer_arn, lb.state_code
             )
-            return None, False, LB_NOT_ACTIVE_RETRY
+            return None, False, self.lb_not_active_retry
 
         logger.info("LoadBalancer is %s", lb.load_balancer_arn)
         accelerators = self.list_global_accelerator_by_resource(
@@ -457,7 +457,7 @@ class GlobalAcceleratorMixin:
             logger.warning(
                 "LoadBalancer %s is not Active: %s", lb.load_balancer_arn, lb.state_code
             )
-            return None, LB_NOT_ACTIVE_RETRY
+            return None, self.lb_not_active_retry
         metrics.observe_aws_call("globalaccelerator", "AddEndpoints")
         added = self.ga.add_endpoints(
             endpoint_group.endpoint_group_arn,

@@ -17,7 +17,7 @@ from . import types as t
 
 logger = logging.getLogger(__name__)
 
-GA_MISSING_RETRY = 60.0  # seconds (reference route53.go:72-76)
+DEFAULT_GA_MISSING_RETRY = 60.0  # seconds (reference route53.go:72-76)
 TXT_TTL = 300  # seconds (reference :276)
 
 
@@ -113,10 +113,10 @@ class Route53Mixin:
         )
         if len(accelerators) > 1:
             logger.error("Too many Global Accelerators for %s", lb_hostname)
-            return False, GA_MISSING_RETRY
+            return False, self.ga_missing_retry
         if not accelerators:
             logger.error("Could not find Global Accelerator for %s", lb_hostname)
-            return False, GA_MISSING_RETRY
+            return False, self.ga_missing_retry
         accelerator = accelerators[0]
 
         owner_value = route53_owner_value(cluster_name, resource, ns, name)

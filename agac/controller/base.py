@@ -1,0 +1,91 @@
+"""Shared controller plumbing: object filters and the worker-thread runner."""
+
+from __future__ import annotations
+
+import logging
+import threading
+from typing import Callable, List
+
+from ..apis import (
+    AWS_GLOBAL_ACCELERATOR_MANAGED_ANNOTATION,
+    AWS_LOAD_BALANCER_TYPE_ANNOTATION,
+    INGRESS_CLASS_ANNOTATION,
+    ROUTE53_HOSTNAME_ANNOTATION,
+)
+from ..apis import core as corev1
+from ..apis.meta import to_dict
+
+logger = logging.getLogger(__name__)
+
+
+# ---------------------------------------------------------------------------
+# Filters (reference ga/service.go:18-26, ga/ingress.go:19-27,
+# ga/controller.go:243-259, r53/controller.go:243-252)
+# ---------------------------------------------------------------------------
+def was_load_balancer_service(svc: corev1.Service) -> bool:
+    """type=LoadBalancer AND (aws-load-balancer-type annotation present OR
+    spec.loadBalancerClass set)."""
+    if svc.spec.type != corev1.SERVICE_TYPE_LOAD_BALANCER:
+        return False
+    return (
+        AWS_LOAD_BALANCER_TYPE_ANNOTATION in svc.metadata.annotations
+        or svc.spec.load_balancer_class is not None
+    )
+
+
+def was_alb_ingress(ingress: corev1.Ingress) -> bool:
+    """spec.ingressClassName == "alb" OR the legacy class annotation exists."""
+    if ingress.spec.ingress_class_name == "alb":
+        return True
+    return INGRESS_CLASS_ANNOTATION in ingress.metadata.annotations
+
+
+def has_managed_annotation(obj) -> bool:
+    return AWS_GLOBAL_ACCELERATOR_MANAGED_ANNOTATION in obj.metadata.annotations
+
+
+def managed_annotation_changed(old, new) -> bool:
+    return (
+        AWS_GLOBAL_ACCELERATOR_MANAGED_ANNOTATION in old.metadata.annotations
+    ) != (AWS_GLOBAL_ACCELERATOR_MANAGED_ANNOTATION in new.metadata.annotations)
+
+
+def has_hostname_annotation(obj) -> bool:
+    return ROUTE53_HOSTNAME_ANNOTATION in obj.metadata.annotations
+
+
+def hostname_annotation_changed(old, new) -> bool:
+    return (ROUTE53_HOSTNAME_ANNOTATION in old.metadata.annotations) != (
+        ROUTE53_HOSTNAME_ANNOTATION in new.metadata.annotations
+    )
+
+
+def objects_equal(old, new) -> bool:
+    """reflect.DeepEqual guard used by every update notification."""
+    return to_dict(old) == to_dict(new)
+
+
+# ---------------------------------------------------------------------------
+# Worker runner
+# ---------------------------------------------------------------------------
+def spawn_workers(
+    threadiness: int, worker_fn: Callable[[], None], name: str, stop: threading.Event
+) -> List[threading.Thread]:
+    """Spawn worker threads that re-enter ``worker_fn`` until stop
+    (go wait.Until(worker, 1s, stopCh), reference ga/controller.go:208-213)."""
+
+    def loop():
+        while not stop.is_set():
+            try:
+                worker_fn()
+                return  # worker_fn exits only on queue shutdown
+            except Exception:
+                logger.exception("worker %s crashed; restarting", name)
+                stop.wait(1.0)
+
+    threads = []
+    for i in range(threadiness):
+        thread = threading.Thread(target=loop, name=f"{name}-worker-{i}", daemon=True)
+        thread.start()
+        threads.append(thread)
+    return threads
