@@ -1,0 +1,178 @@
+"""Command-line interface.
+
+click-based equivalent of the reference's cobra CLI (``cmd/``):
+
+- ``agac controller`` — start the controller manager under leader election
+  (flags: --workers/-w, --cluster-name/-c, --kubeconfig, --master,
+  reference ``cmd/controller/controller.go:24-98``);
+- ``agac webhook``   — start the admission webhook server (flags:
+  --tls-cert-file, --tls-private-key-file, --port, --ssl,
+  reference ``cmd/webhook/webhook.go:17-41``);
+- ``agac version``   — print version info (reference ``cmd/version.go``).
+
+kubeconfig resolution order matches viper wiring: flag → $KUBECONFIG →
+~/.kube/config.  ``POD_NAMESPACE`` selects the leader-election namespace.
+The kube backend is selected by --api: "memory" runs against an embedded
+in-process API store (with the in-memory AWS fake — demo/e2e mode), "http"
+connects to an agac API server (see ``agac.kube.rest``).
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import sys
+
+import click
+
+from . import __version__
+
+logger = logging.getLogger(__name__)
+
+
+def _setup_logging(verbosity: int):
+    level = logging.WARNING
+    if verbosity == 1:
+        level = logging.INFO
+    elif verbosity >= 2:
+        level = logging.DEBUG
+    logging.basicConfig(
+        level=level,
+        format="%(asctime)s %(levelname).1s %(name)s %(message)s",
+        stream=sys.stderr,
+    )
+
+
+@click.group()
+@click.option("-v", "--verbose", count=True, help="Increase log verbosity (klog-style).")
+def cli(verbose: int):
+    """aws-global-accelerator-controller (agac)."""
+    _setup_logging(verbose)
+
+
+def resolve_kubeconfig(kubeconfig: str) -> str:
+    """flag → $KUBECONFIG → ~/.kube/config (reference controller.go:84-98)."""
+    if kubeconfig:
+        return kubeconfig
+    env = os.environ.get("KUBECONFIG", "")
+    if env:
+        return env
+    default = os.path.expanduser("~/.kube/config")
+    if os.path.exists(default):
+        return default
+    return ""
+
+
+@cli.command()
+@click.option("-w", "--workers", default=1, show_default=True, help="Concurrent workers number for controller.")
+@click.option("-c", "--cluster-name", default="default", show_default=True, help="Owner cluster name which is used in resource tags.")
+@click.option("--kubeconfig", default="", help="Path to a kubeconfig. Only required if out-of-cluster.")
+@click.option("--master", default="", help="The address of the Kubernetes API server. Overrides any value in kubeconfig.")
+@click.option("--api", type=click.Choice(["memory", "http"]), default="memory", show_default=True, help="Kube API backend: embedded in-memory store or an HTTP API server.")
+@click.option("--metrics-port", default=0, help="Serve Prometheus metrics on this port (0 = disabled).")
+@click.option("--leader-elect/--no-leader-elect", default=True, show_default=True)
+def controller(workers, cluster_name, kubeconfig, master, api, metrics_port, leader_elect):
+    """Start controller."""
+    from .controller.endpointgroupbinding import EndpointGroupBindingConfig
+    from .controller.globalaccelerator import GlobalAcceleratorConfig
+    from .controller.route53 import Route53Config
+    from .kube.leaderelection import LeaderElector
+    from .manager import ControllerConfig, Manager
+    from .metrics import start_metrics_server
+    from .signals import setup_signal_handler
+
+    if api == "memory":
+        from .cloudprovider.aws.client import FakeCloudFactory
+        from .kube.client import InMemoryKubeClient
+
+        kube_client = InMemoryKubeClient()
+        cloud_factory = FakeCloudFactory()
+        logger.info("Using embedded in-memory API store + AWS fake")
+    else:
+        from .kube.rest import RestKubeClient
+
+        server = master or os.environ.get("AGAC_API_SERVER", "")
+        if not server:
+            raise click.UsageError("--api http requires --master or $AGAC_API_SERVER")
+        kube_client = RestKubeClient(server)
+        try:
+            from .cloudprovider.aws.client import boto3_cloud_factory
+
+            cloud_factory = boto3_cloud_factory()
+        except RuntimeError:
+            from .cloudprovider.aws.client import FakeCloudFactory
+
+            logger.warning("boto3 unavailable; falling back to the in-memory AWS fake")
+            cloud_factory = FakeCloudFactory()
+        kc = resolve_kubeconfig(kubeconfig)
+        if kc:
+            logger.info("Using kubeconfig: %s", kc)
+
+    if metrics_port:
+        start_metrics_server(metrics_port)
+
+    namespace = os.environ.get("POD_NAMESPACE", "default")
+    config = ControllerConfig(
+        global_accelerator=GlobalAcceleratorConfig(workers=workers, cluster_name=cluster_name),
+        route53=Route53Config(workers=workers, cluster_name=cluster_name),
+        endpoint_group_binding=EndpointGroupBindingConfig(workers=workers),
+    )
+    stop = setup_signal_handler()
+
+    def run_manager(stop_leading):
+        manager = Manager()
+        manager.run(kube_client, config, cloud_factory, stop_leading, block=True)
+
+    if leader_elect:
+        elector = LeaderElector(
+            kube_client,
+            name="aws-global-accelerator-controller",
+            namespace=namespace,
+            on_started_leading=run_manager,
+            on_stopped_leading=lambda: logger.info("leadership ended"),
+            on_new_leader=lambda ident: logger.info("new leader elected: %s", ident),
+        )
+        logger.info("leader election id: %s", elector.identity)
+        elector.run(stop)
+        # reference exits 0 after losing/releasing the lease
+        sys.exit(0)
+    else:
+        run_manager(stop)
+
+
+@cli.command()
+@click.option("--tls-cert-file", default="", help="TLS certificate file path.")
+@click.option("--tls-private-key-file", default="", help="TLS private key file path.")
+@click.option("--port", default=8443, show_default=True, help="Listen port.")
+@click.option("--ssl/--no-ssl", default=True, show_default=True, help="Enable TLS.")
+def webhook(tls_cert_file, tls_private_key_file, port, ssl):
+    """Start validating admission webhook server."""
+    from .webhook.server import serve
+
+    if not ssl:
+        tls_cert_file = tls_private_key_file = ""
+    serve(port, tls_cert_file, tls_private_key_file)
+
+
+@cli.command()
+def version():
+    """Print version."""
+    click.echo(f"aws-global-accelerator-controller (agac) {__version__}")
+
+
+@cli.command("apiserver")
+@click.option("--port", default=8001, show_default=True, help="Listen port.")
+def apiserver(port):
+    """Serve the in-memory API store over HTTP (hermetic e2e backend)."""
+    from .kube.httpapi import serve_store
+    from .kube.store import APIStore
+
+    serve_store(APIStore(), port)
+
+
+def main():
+    cli()
+
+
+if __name__ == "__main__":
+    main()

@@ -89,3 +89,19 @@ def spawn_workers(
         thread.start()
         threads.append(thread)
     return threads
+
+
+def make_queue_rate_limiter(qps: float, burst: int):
+    """Controller queue limiter: per-item exponential backoff + overall
+    token bucket (client-go DefaultControllerRateLimiter shape) with
+    configurable qps/burst — the reference hardcodes 10 qps / 100 burst,
+    which caps sustained reconcile throughput at 10 objects/s."""
+    from ..kube.workqueue import (
+        BucketRateLimiter,
+        ItemExponentialFailureRateLimiter,
+        MaxOfRateLimiter,
+    )
+
+    return MaxOfRateLimiter(
+        ItemExponentialFailureRateLimiter(), BucketRateLimiter(qps=qps, burst=burst)
+    )

@@ -33,7 +33,7 @@ from ..kube.events import EventRecorder
 from ..kube.informer import wait_for_cache_sync
 from ..kube.store import is_not_found
 from ..kube.workqueue import RateLimitingQueue
-from .base import spawn_workers
+from .base import make_queue_rate_limiter, spawn_workers
 
 logger = logging.getLogger(__name__)
 
@@ -43,6 +43,8 @@ CONTROLLER_AGENT_NAME = "endpoint-group-binding-controller"
 @dataclass
 class EndpointGroupBindingConfig:
     workers: int = 1
+    queue_qps: float = 10.0
+    queue_burst: int = 100
 
 
 class EndpointGroupBindingController:
@@ -54,7 +56,10 @@ class EndpointGroupBindingController:
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
-        self.workqueue = RateLimitingQueue(name="EndpointGroupBinding")
+        self.workqueue = RateLimitingQueue(
+            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
+            name="EndpointGroupBinding",
+        )
 
         self.service_informer = informer_factory.services()
         self.service_lister = self.service_informer.lister()

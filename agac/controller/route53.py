@@ -27,6 +27,7 @@ from ..kube.events import EventRecorder
 from ..kube.informer import wait_for_cache_sync
 from ..kube.workqueue import RateLimitingQueue
 from .base import (
+    make_queue_rate_limiter,
     has_hostname_annotation,
     hostname_annotation_changed,
     objects_equal,
@@ -43,6 +44,9 @@ CONTROLLER_AGENT_NAME = "route53-controller"
 class Route53Config:
     workers: int = 1
     cluster_name: str = "default"
+    # queue token-bucket rate (client-go default 10/100); raise for scale
+    queue_qps: float = 10.0
+    queue_burst: int = 100
 
 
 class Route53Controller:
@@ -51,8 +55,14 @@ class Route53Controller:
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
-        self.service_queue = RateLimitingQueue(name=CONTROLLER_AGENT_NAME + "-service")
-        self.ingress_queue = RateLimitingQueue(name=CONTROLLER_AGENT_NAME + "-ingress")
+        self.service_queue = RateLimitingQueue(
+            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
+            name=CONTROLLER_AGENT_NAME + "-service",
+        )
+        self.ingress_queue = RateLimitingQueue(
+            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
+            name=CONTROLLER_AGENT_NAME + "-ingress",
+        )
 
         service_informer = informer_factory.services()
         self.service_lister = service_informer.lister()

@@ -1,0 +1,234 @@
+#!/usr/bin/env python3
+"""Benchmark: reconcile throughput of the full controller stack.
+
+BASELINE.json re-tiers this repo to infra/k8s-controller (the reference is a
+network-I/O-bound Go Kubernetes controller with no published benchmarks and
+no ML surface) and names the proxy metric: **reconcile latency (event →
+converged cloud state)**.  This bench measures exactly that, end to end:
+
+Each rank runs the complete framework — in-memory kube API server, shared
+informers, three controllers (GlobalAccelerator / Route53 /
+EndpointGroupBinding) with worker threads and rate-limited queues, and the
+stateful in-memory AWS fake.  One *step* mutates every synthetic Service
+(port flip) and blocks until the fake AWS shows every Global Accelerator
+listener converged to the new spec, i.e. steps time the full
+watch → informer → queue → reconcile → cloud-API pipeline.
+
+value = objects converged per second, aggregated over all ranks
+(weak scaling: each rank owns an independent controller stack of
+--objects objects; a k8s controller has no inter-rank communication, so
+ranks only synchronize for timing).
+
+Contract: rank 0 prints exactly one JSON line.  Works single-process
+(default) and under torch.distributed.run with one rank per GPU (the
+workload is CPU-bound; GPUs are intentionally unused — tier mismatch per
+BASELINE.json).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import threading
+import time
+
+
+def build_stack(objects: int, workers: int):
+    from agac.apis import core as corev1
+    from agac.apis.meta import ObjectMeta
+    from agac.cloudprovider.aws.client import FakeCloudFactory
+    from agac.cloudprovider.fake import FakeAWSBackend
+    from agac.controller.endpointgroupbinding import EndpointGroupBindingConfig
+    from agac.controller.globalaccelerator import GlobalAcceleratorConfig
+    from agac.controller.route53 import Route53Config
+    from agac.kube.client import InMemoryKubeClient
+    from agac.manager import ControllerConfig, Manager
+
+    MANAGED = (
+        "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-managed"
+    )
+    LB_TYPE = "service.beta.kubernetes.io/aws-load-balancer-type"
+    region = "us-east-1"
+
+    client = InMemoryKubeClient()
+    backend = FakeAWSBackend(deploy_after_describes=0)
+    factory = FakeCloudFactory(backend)
+    stop = threading.Event()
+    manager = Manager()
+    qps = 1e9  # measure the framework, not the client-go anti-thundering-herd default
+    config = ControllerConfig(
+        global_accelerator=GlobalAcceleratorConfig(
+            workers=workers, queue_qps=qps, queue_burst=1 << 30
+        ),
+        route53=Route53Config(workers=workers, queue_qps=qps, queue_burst=1 << 30),
+        endpoint_group_binding=EndpointGroupBindingConfig(
+            workers=workers, queue_qps=qps, queue_burst=1 << 30
+        ),
+    )
+    manager.run(client, config, factory, stop, resync_period=300.0, block=False)
+    if not manager.wait_until_ready():
+        raise RuntimeError("controllers did not become ready")
+
+    services = []
+    for i in range(objects):
+        lb = backend.elbv2.create_load_balancer(f"lb-{i}", region=region)
+        svc = corev1.Service(
+            metadata=ObjectMeta(
+                name=f"svc-{i}",
+                namespace="default",
+                annotations={LB_TYPE: "nlb", MANAGED: "true"},
+            ),
+            spec=corev1.ServiceSpec(
+                type="LoadBalancer",
+                ports=[corev1.ServicePort(port=80, protocol="TCP")],
+            ),
+            status=corev1.ServiceStatus(
+                load_balancer=corev1.LoadBalancerStatus(
+                    ingress=[corev1.LoadBalancerIngress(hostname=lb.dns_name)]
+                )
+            ),
+        )
+        client.create(svc)
+        services.append(svc.metadata.name)
+
+    return client, backend, services, stop
+
+
+def converged(backend, owner_to_port: dict) -> bool:
+    """True when every managed accelerator's listener carries the port its
+    owning service currently specifies."""
+    ga = backend.ga
+    with backend.lock:
+        tags_by_arn = ga._tags
+        seen = 0
+        for arn, tags in tags_by_arn.items():
+            owner = tags.get("aws-global-accelerator-owner")
+            want = owner_to_port.get(owner)
+            if want is None:
+                continue
+            listeners = [
+                l for l, a in ga._listener_owner.items() if a == arn
+            ]
+            if len(listeners) != 1:
+                return False
+            ports = [p.from_port for p in ga._listeners[listeners[0]].port_ranges]
+            if ports != [want]:
+                return False
+            seen += 1
+        return seen == len(owner_to_port)
+
+
+def run_step(client, backend, services, step_idx: int, timeout: float = 120.0):
+    """Mutate every service's port and wait for full convergence."""
+    port = 8000 + (step_idx % 2)
+    owner_to_port = {}
+    for name in services:
+        svc = client.get("Service", "default", name)
+        svc.spec.ports[0].port = port
+        client.update(svc)
+        owner_to_port[f"service/default/{name}"] = port
+    deadline = time.monotonic() + timeout
+    while not converged(backend, owner_to_port):
+        if time.monotonic() >= deadline:
+            raise TimeoutError(f"step {step_idx} did not converge in {timeout}s")
+        time.sleep(0.001)
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=10)
+    parser.add_argument("--warmup", type=int, default=3)
+    parser.add_argument("--objects", type=int, default=64)
+    parser.add_argument("--workers", type=int, default=8)
+    args = parser.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist  # noqa: PLC0415
+
+        # the workload is CPU-bound (k8s controller); gloo synchronizes
+        # timing across the per-GPU ranks without touching the GPUs
+        dist.init_process_group(backend="gloo")
+
+    client, backend, services, stop = build_stack(args.objects, args.workers)
+
+    try:
+        # initial creation converges during warmup setup
+        owner_to_port = {f"service/default/{n}": 80 for n in services}
+        deadline = time.monotonic() + 120.0
+        while not converged(backend, owner_to_port):
+            if time.monotonic() >= deadline:
+                raise TimeoutError("initial convergence timed out")
+            time.sleep(0.001)
+
+        for w in range(args.warmup):
+            run_step(client, backend, services, w)
+
+        try:
+            import torch
+
+            cuda = torch.cuda.is_available()
+        except ImportError:
+            torch, cuda = None, False
+
+        if dist is not None:
+            dist.barrier()
+        if cuda:
+            torch.cuda.synchronize()
+        start = time.monotonic()
+        for k in range(args.steps):
+            run_step(client, backend, services, args.warmup + k)
+        if cuda:
+            torch.cuda.synchronize()
+        elapsed = time.monotonic() - start
+        if dist is not None:
+            import torch as _t
+
+            t = _t.tensor([elapsed], dtype=_t.float64)
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            dist.barrier()
+            elapsed = float(t.item())
+
+        ms_per_step = elapsed / args.steps * 1000.0
+        # whole-job aggregate: every rank converged `objects` objects per step
+        value = args.objects * args.steps * world_size / elapsed
+
+        if rank == 0:
+            print(
+                json.dumps(
+                    {
+                        "metric": "reconciles_per_s",
+                        "value": round(value, 2),
+                        "unit": "objects_converged/s",
+                        "n_gpus": world_size,
+                        "steps": args.steps,
+                        "warmup": args.warmup,
+                        "ms_per_step": round(ms_per_step, 3),
+                        "higher_is_better": True,
+                        "scaling": "weak",
+                        "vs_baseline": None,
+                        "dtype": "n/a",
+                        "data": "synthetic",
+                        "config": {
+                            "model": "k8s-controller reconcile loop (BASELINE.json: tier-mismatch, no ML model; proxy metric = reconcile latency event->converged)",
+                            "objects_per_rank": args.objects,
+                            "workers_per_queue": args.workers,
+                            "parallelism": f"independent controller stack per rank (x{world_size})",
+                        },
+                    }
+                )
+            )
+            sys.stdout.flush()
+    finally:
+        stop.set()
+        if dist is not None:
+            dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
