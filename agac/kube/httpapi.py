@@ -1,0 +1,233 @@
+"""HTTP API server over the in-memory store.
+
+The kube-apiserver stand-in for e2e testing with a real network boundary
+(the analogue of the reference's kind-based e2e tier, ``e2e/``): serves the
+``APIStore`` over HTTP with list/get/create/update/update-status/delete and
+chunked-streaming watch, so ``RestKubeClient`` + the controller manager run
+against it exactly as they would against a remote API server.
+
+Wire scheme (generic, k8s-shaped):
+
+- ``GET    /apis/{kind}[?namespace=ns]``        → {"items": [...], "resourceVersion": N}
+- ``GET    /apis/{kind}/{ns}/{name}``           → object
+- ``POST   /apis/{kind}``                        → created object
+- ``PUT    /apis/{kind}/{ns}/{name}``           → updated object
+- ``PUT    /apis/{kind}/{ns}/{name}/status``    → updated object
+- ``DELETE /apis/{kind}/{ns}/{name}``           → {}
+- ``GET    /watch/{kind}[?namespace=&resourceVersion=N]``
+      → ndjson stream of {"type", "object", "resourceVersion"}
+- ``GET    /healthz``
+
+Errors: JSON ``{"code", "reason", "message"}`` with the matching HTTP
+status; reasons NotFound / AlreadyExists / Conflict / Gone map back to the
+typed store errors in the client.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import threading
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from urllib.parse import parse_qs, urlparse
+
+from ..apis.meta import from_dict, to_dict
+from .client import class_for_kind
+from .store import (
+    AlreadyExistsError,
+    APIError,
+    APIStore,
+    ConflictError,
+    GoneError,
+    NotFoundError,
+)
+
+logger = logging.getLogger(__name__)
+
+
+def _error_body(e: APIError) -> dict:
+    reason = {
+        NotFoundError: "NotFound",
+        AlreadyExistsError: "AlreadyExists",
+        ConflictError: "Conflict",
+        GoneError: "Gone",
+    }.get(type(e), "InternalError")
+    return {"code": e.code, "reason": reason, "message": str(e)}
+
+
+class _Handler(BaseHTTPRequestHandler):
+    protocol_version = "HTTP/1.1"
+    store: APIStore = None  # set by server factory
+
+    def log_message(self, fmt, *args):  # noqa: A003
+        logger.debug(fmt, *args)
+
+    # -- helpers -----------------------------------------------------------
+    def _json(self, code: int, payload: dict):
+        body = json.dumps(payload).encode()
+        self.send_response(code)
+        self.send_header("Content-Type", "application/json")
+        self.send_header("Content-Length", str(len(body)))
+        self.end_headers()
+        self.wfile.write(body)
+
+    def _api_error(self, e: APIError):
+        self._json(e.code, _error_body(e))
+
+    def _obj_with_kind(self, obj) -> dict:
+        d = to_dict(obj)
+        d["kind"] = type(obj).kind
+        d["apiVersion"] = type(obj).api_version
+        return d
+
+    def _read_body(self):
+        length = int(self.headers.get("Content-Length") or 0)
+        return json.loads(self.rfile.read(length)) if length else {}
+
+    def _route(self):
+        parsed = urlparse(self.path)
+        parts = [p for p in parsed.path.split("/") if p]
+        query = {k: v[0] for k, v in parse_qs(parsed.query).items()}
+        return parts, query
+
+    # -- verbs -------------------------------------------------------------
+    def do_GET(self):  # noqa: N802
+        parts, query = self._route()
+        try:
+            if parts == ["healthz"]:
+                self._json(200, {"status": "ok"})
+            elif len(parts) == 2 and parts[0] == "apis":
+                items, rv = self.store.list(parts[1], query.get("namespace"))
+                self._json(
+                    200,
+                    {"items": [self._obj_with_kind(o) for o in items], "resourceVersion": rv},
+                )
+            elif len(parts) == 4 and parts[0] == "apis":
+                obj = self.store.get(parts[1], parts[2], parts[3])
+                self._json(200, self._obj_with_kind(obj))
+            elif len(parts) == 2 and parts[0] == "watch":
+                self._serve_watch(parts[1], query)
+            else:
+                self._json(404, {"code": 404, "reason": "NotFound", "message": "no such route"})
+        except APIError as e:
+            self._api_error(e)
+        except BrokenPipeError:
+            pass
+
+    def do_POST(self):  # noqa: N802
+        parts, _ = self._route()
+        try:
+            if len(parts) == 2 and parts[0] == "apis":
+                cls = class_for_kind(parts[1])
+                obj = from_dict(cls, self._read_body())
+                created = self.store.create(obj)
+                self._json(201, self._obj_with_kind(created))
+            else:
+                self._json(404, {"code": 404, "reason": "NotFound", "message": "no such route"})
+        except APIError as e:
+            self._api_error(e)
+
+    def do_PUT(self):  # noqa: N802
+        parts, _ = self._route()
+        try:
+            if len(parts) in (4, 5) and parts[0] == "apis":
+                kind, ns, name = parts[1], parts[2], parts[3]
+                cls = class_for_kind(kind)
+                obj = from_dict(cls, self._read_body())
+                obj.metadata.namespace = ns
+                obj.metadata.name = name
+                if len(parts) == 5 and parts[4] == "status":
+                    updated = self.store.update_status(obj)
+                else:
+                    updated = self.store.update(obj)
+                self._json(200, self._obj_with_kind(updated))
+            else:
+                self._json(404, {"code": 404, "reason": "NotFound", "message": "no such route"})
+        except APIError as e:
+            self._api_error(e)
+
+    def do_DELETE(self):  # noqa: N802
+        parts, _ = self._route()
+        try:
+            if len(parts) == 4 and parts[0] == "apis":
+                self.store.delete(parts[1], parts[2], parts[3])
+                self._json(200, {})
+            else:
+                self._json(404, {"code": 404, "reason": "NotFound", "message": "no such route"})
+        except APIError as e:
+            self._api_error(e)
+
+    # -- watch streaming ----------------------------------------------------
+    def _serve_watch(self, kind: str, query: dict):
+        namespace = query.get("namespace")
+        rv = query.get("resourceVersion")
+        watch = self.store.watch(
+            kind, namespace, int(rv) if rv is not None else None
+        )
+        self.send_response(200)
+        self.send_header("Content-Type", "application/json")
+        self.send_header("Transfer-Encoding", "chunked")
+        self.end_headers()
+        try:
+            while True:
+                event = watch.get(timeout=5.0)
+                if event is None:
+                    # heartbeat keeps half-open connections detectable
+                    self._write_chunk(b"")
+                    continue
+                line = json.dumps(
+                    {
+                        "type": event.type,
+                        "object": self._obj_with_kind(event.obj),
+                        "resourceVersion": event.resource_version,
+                    }
+                ).encode() + b"\n"
+                self._write_chunk(line)
+        except (BrokenPipeError, ConnectionResetError, OSError):
+            pass
+        finally:
+            watch.stop()
+
+    def _write_chunk(self, data: bytes):
+        if not data:
+            # zero-length would terminate chunked encoding; send a newline
+            data = b"\n"
+        self.wfile.write(f"{len(data):x}\r\n".encode() + data + b"\r\n")
+        self.wfile.flush()
+
+
+class APIServer:
+    """Owns the HTTP listener for one APIStore."""
+
+    def __init__(self, store: APIStore, port: int = 0, host: str = "127.0.0.1"):
+        handler = type("BoundHandler", (_Handler,), {"store": store})
+        self.httpd = ThreadingHTTPServer((host, port), handler)
+        self.httpd.daemon_threads = True
+        self.store = store
+
+    @property
+    def port(self) -> int:
+        return self.httpd.server_address[1]
+
+    @property
+    def url(self) -> str:
+        host, port = self.httpd.server_address[:2]
+        return f"http://{host}:{port}"
+
+    def start(self) -> threading.Thread:
+        thread = threading.Thread(
+            target=self.httpd.serve_forever, name="agac-apiserver", daemon=True
+        )
+        thread.start()
+        return thread
+
+    def shutdown(self):
+        self.httpd.shutdown()
+        self.httpd.server_close()
+
+
+def serve_store(store: APIStore, port: int):
+    """Blocking CLI entry point."""
+    server = APIServer(store, port, host="")
+    logger.info("API server listening on :%d", server.port)
+    server.httpd.serve_forever()
