@@ -503,9 +503,9 @@ class FakeAWSBackend:
 
 
 def _copy(obj):
-    import copy
+    from ...apis.meta import deep_copy
 
-    return copy.deepcopy(obj)
+    return deep_copy(obj)
 
 
 def _to_description(config: t.EndpointConfiguration) -> t.EndpointDescription:

@@ -61,7 +61,15 @@ def hostname_annotation_changed(old, new) -> bool:
 
 
 def objects_equal(old, new) -> bool:
-    """reflect.DeepEqual guard used by every update notification."""
+    """reflect.DeepEqual guard used by every update notification.
+
+    Informer-delivered pairs are equal exactly when they carry the same
+    resourceVersion (a resync re-delivers the identical cached object —
+    which client-go's DeepEqual also skips, since UpdateFunc gets the same
+    pointer twice).  O(1) instead of serializing both objects."""
+    old_rv = old.metadata.resource_version
+    if old_rv and old_rv == new.metadata.resource_version:
+        return True
     return to_dict(old) == to_dict(new)
 
 

@@ -222,9 +222,7 @@ class APIStore:
             stored.metadata.generation = existing.metadata.generation
             if hasattr(stored, "status"):
                 stored.status = metalib.deep_copy(existing.status)
-            if hasattr(stored, "spec") and metalib.to_dict(stored.spec) != metalib.to_dict(
-                existing.spec
-            ):
+            if hasattr(stored, "spec") and stored.spec != existing.spec:
                 stored.metadata.generation += 1
 
             if (

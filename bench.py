@@ -98,23 +98,23 @@ def build_stack(objects: int, workers: int):
 
 def converged(backend, owner_to_port: dict) -> bool:
     """True when every managed accelerator's listener carries the port its
-    owning service currently specifies."""
+    owning service currently specifies.  Single O(N) pass under the lock so
+    the convergence poll doesn't stall reconcile workers."""
     ga = backend.ga
     with backend.lock:
-        tags_by_arn = ga._tags
+        listeners_by_acc = {}
+        for listener_arn, acc_arn in ga._listener_owner.items():
+            listeners_by_acc.setdefault(acc_arn, []).append(listener_arn)
         seen = 0
-        for arn, tags in tags_by_arn.items():
-            owner = tags.get("aws-global-accelerator-owner")
-            want = owner_to_port.get(owner)
+        for arn, tags in ga._tags.items():
+            want = owner_to_port.get(tags.get("aws-global-accelerator-owner"))
             if want is None:
                 continue
-            listeners = [
-                l for l, a in ga._listener_owner.items() if a == arn
-            ]
+            listeners = listeners_by_acc.get(arn, ())
             if len(listeners) != 1:
                 return False
-            ports = [p.from_port for p in ga._listeners[listeners[0]].port_ranges]
-            if ports != [want]:
+            ranges = ga._listeners[listeners[0]].port_ranges
+            if len(ranges) != 1 or ranges[0].from_port != want:
                 return False
             seen += 1
         return seen == len(owner_to_port)
