@@ -1,0 +1,75 @@
+"""Event recorder aggregation + Prometheus metrics hooks."""
+
+from agac.apis import core as corev1
+from agac.apis.meta import ObjectMeta
+from agac.kube.client import InMemoryKubeClient
+from agac.kube.events import EventRecorder
+
+
+def mk_obj():
+    return corev1.Service(metadata=ObjectMeta(name="web", namespace="default", uid="u1"))
+
+
+def list_events(client):
+    items, _ = client.list("Event")
+    return items
+
+
+def test_event_creates_object_with_reference():
+    client = InMemoryKubeClient()
+    recorder = EventRecorder(client, "test-controller")
+    recorder.event(mk_obj(), "Normal", "GlobalAcceleratorCreated", "created arn:x")
+    events = list_events(client)
+    assert len(events) == 1
+    ev = events[0]
+    assert ev.reason == "GlobalAcceleratorCreated"
+    assert ev.type == "Normal"
+    assert ev.involved_object.kind == "Service"
+    assert ev.involved_object.name == "web"
+    assert ev.source.component == "test-controller"
+    assert ev.count == 1
+
+
+def test_repeated_event_aggregates_count():
+    client = InMemoryKubeClient()
+    recorder = EventRecorder(client, "c")
+    for _ in range(3):
+        recorder.event(mk_obj(), "Normal", "Reason", "same message")
+    events = list_events(client)
+    assert len(events) == 1
+    assert events[0].count == 3
+
+
+def test_different_messages_make_distinct_events():
+    client = InMemoryKubeClient()
+    recorder = EventRecorder(client, "c")
+    recorder.eventf(mk_obj(), "Normal", "Reason", "msg %d", 1)
+    recorder.eventf(mk_obj(), "Normal", "Reason", "msg %d", 2)
+    assert len(list_events(client)) == 2
+
+
+def test_recorder_never_raises():
+    class BrokenClient(InMemoryKubeClient):
+        def create(self, obj):
+            raise RuntimeError("apiserver down")
+
+    recorder = EventRecorder(BrokenClient(), "c")
+    recorder.event(mk_obj(), "Normal", "Reason", "m")  # must not raise
+
+
+def test_metrics_counters_observable():
+    from prometheus_client import REGISTRY
+
+    from agac import metrics
+
+    metrics.observe_reconcile("test-queue-xyz", "success", 0.01)
+    metrics.observe_aws_call("elbv2-test", "DescribeLoadBalancers")
+    value = REGISTRY.get_sample_value(
+        "agac_reconcile_total", {"queue": "test-queue-xyz", "outcome": "success"}
+    )
+    assert value == 1.0
+    value = REGISTRY.get_sample_value(
+        "agac_aws_api_calls_total",
+        {"service": "elbv2-test", "operation": "DescribeLoadBalancers"},
+    )
+    assert value == 1.0
