@@ -90,6 +90,8 @@ class EndpointGroupBindingController:
         if not wait_for_cache_sync(
             stop, self.binding_informer, self.service_informer, self.ingress_informer
         ):
+            if stop.is_set():
+                return  # shutdown requested before caches synced
             raise RuntimeError("failed to wait for caches to sync")
         spawn_workers(threadiness, self._run_worker, CONTROLLER_AGENT_NAME, stop)
         stop.wait()

@@ -125,6 +125,8 @@ class GlobalAcceleratorController:
     def run(self, threadiness: int, stop: threading.Event):
         logger.info("Starting GlobalAccelerator controller")
         if not wait_for_cache_sync(stop, self.service_informer, self.ingress_informer):
+            if stop.is_set():
+                return  # shutdown requested before caches synced
             raise RuntimeError("failed to wait for caches to sync")
         logger.info("Starting workers")
         spawn_workers(threadiness, self._run_service_worker, CONTROLLER_AGENT_NAME + "-service", stop)

@@ -121,6 +121,8 @@ class Route53Controller:
     def run(self, threadiness: int, stop: threading.Event):
         logger.info("Starting Route53 controller")
         if not wait_for_cache_sync(stop, self.service_informer, self.ingress_informer):
+            if stop.is_set():
+                return  # shutdown requested before caches synced
             raise RuntimeError("failed to wait for caches to sync")
         spawn_workers(threadiness, self._run_service_worker, CONTROLLER_AGENT_NAME + "-service", stop)
         spawn_workers(threadiness, self._run_ingress_worker, CONTROLLER_AGENT_NAME + "-ingress", stop)
