@@ -46,11 +46,14 @@ logger = logging.getLogger(__name__)
 
 
 def _error_body(e: APIError) -> dict:
+    from .admission import AdmissionDeniedError
+
     reason = {
         NotFoundError: "NotFound",
         AlreadyExistsError: "AlreadyExists",
         ConflictError: "Conflict",
         GoneError: "Gone",
+        AdmissionDeniedError: "Forbidden",
     }.get(type(e), "InternalError")
     return {"code": e.code, "reason": reason, "message": str(e)}
 

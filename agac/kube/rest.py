@@ -28,11 +28,14 @@ from .store import (
 
 logger = logging.getLogger(__name__)
 
+from .admission import AdmissionDeniedError
+
 _REASON_TO_ERROR = {
     "NotFound": NotFoundError,
     "AlreadyExists": AlreadyExistsError,
     "Conflict": ConflictError,
     "Gone": GoneError,
+    "Forbidden": AdmissionDeniedError,
 }
 
 

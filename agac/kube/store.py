@@ -130,6 +130,20 @@ class APIStore:
         self._objects: Dict[str, Dict[Tuple[str, str], object]] = {}
         self._watches: List[_Watch] = []
         self._event_log: deque = deque(maxlen=_EVENT_LOG_SIZE)
+        # registered admission webhooks (agac.kube.admission), consulted
+        # before create/update/delete commits — the apiserver side of
+        # ValidatingWebhookConfiguration
+        self.admission_webhooks: List = []
+
+    def _admit(self, kind: str, operation: str, old, new):
+        """Run admission outside the store lock (webhook calls may do
+        network I/O); commit-time rv checks still serialize writers."""
+        if not self.admission_webhooks:
+            return
+        old_d = metalib.to_dict(old) if old is not None else None
+        new_d = metalib.to_dict(new) if new is not None else None
+        for hook in self.admission_webhooks:
+            hook.admit(kind, operation, old_d, new_d)
 
     # -- helpers -----------------------------------------------------------
     def _bucket(self, kind: str) -> Dict[Tuple[str, str], object]:
@@ -153,6 +167,7 @@ class APIStore:
     # -- CRUD --------------------------------------------------------------
     def create(self, obj):
         kind = type(obj).kind
+        self._admit(kind, "CREATE", None, obj)
         with self._lock:
             key = (obj.metadata.namespace, obj.metadata.name)
             bucket = self._bucket(kind)
@@ -206,6 +221,11 @@ class APIStore:
         """Update spec+metadata.  Bumps generation if spec changed; removes the
         object if it has a deletionTimestamp and finalizers became empty."""
         kind = type(obj).kind
+        try:
+            current = self.get(kind, obj.metadata.namespace, obj.metadata.name)
+        except NotFoundError:
+            current = None
+        self._admit(kind, "UPDATE", current, obj)
         with self._lock:
             key = (obj.metadata.namespace, obj.metadata.name)
             bucket = self._bucket(kind)
@@ -261,6 +281,11 @@ class APIStore:
 
     def delete(self, kind: str, namespace: str, name: str):
         """Finalizer-aware delete (kube-apiserver graceful deletion)."""
+        try:
+            current = self.get(kind, namespace, name)
+        except NotFoundError:
+            current = None
+        self._admit(kind, "DELETE", current, None)
         with self._lock:
             key = (namespace, name)
             bucket = self._bucket(kind)
