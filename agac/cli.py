@@ -156,8 +156,24 @@ def webhook(tls_cert_file, tls_private_key_file, port, ssl):
 
 @cli.command()
 def version():
-    """Print version."""
+    """Print version/revision/build (reference cmd/version.go injects these
+    via -ldflags; here revision/build come from the environment or git)."""
+    import subprocess
+
+    revision = os.environ.get("AGAC_REVISION", "")
+    if not revision:
+        try:
+            revision = subprocess.run(
+                ["git", "rev-parse", "--short", "HEAD"],
+                capture_output=True, text=True, timeout=5,
+                cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+            ).stdout.strip() or "unknown"
+        except Exception:
+            revision = "unknown"
+    build = os.environ.get("AGAC_BUILD", "source")
     click.echo(f"aws-global-accelerator-controller (agac) {__version__}")
+    click.echo(f"revision: {revision}")
+    click.echo(f"build: {build}")
 
 
 @cli.command("apiserver")
