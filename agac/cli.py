@@ -68,7 +68,7 @@ def resolve_kubeconfig(kubeconfig: str) -> str:
 @click.option("-c", "--cluster-name", default="default", show_default=True, help="Owner cluster name which is used in resource tags.")
 @click.option("--kubeconfig", default="", help="Path to a kubeconfig. Only required if out-of-cluster.")
 @click.option("--master", default="", help="The address of the Kubernetes API server. Overrides any value in kubeconfig.")
-@click.option("--api", type=click.Choice(["memory", "http"]), default="memory", show_default=True, help="Kube API backend: embedded in-memory store or an HTTP API server.")
+@click.option("--api", type=click.Choice(["memory", "http", "k8s"]), default="memory", show_default=True, help="Kube API backend: embedded in-memory store, an agac HTTP API server, or a real Kubernetes API server (kubeconfig/in-cluster auth).")
 @click.option("--metrics-port", default=0, help="Serve Prometheus metrics on this port (0 = disabled).")
 @click.option("--leader-elect/--no-leader-elect", default=True, show_default=True)
 def controller(workers, cluster_name, kubeconfig, master, api, metrics_port, leader_elect):
@@ -89,12 +89,23 @@ def controller(workers, cluster_name, kubeconfig, master, api, metrics_port, lea
         cloud_factory = FakeCloudFactory()
         logger.info("Using embedded in-memory API store + AWS fake")
     else:
-        from .kube.rest import RestKubeClient
+        if api == "k8s":
+            from .kube.k8s import K8sKubeClient
+            from .kube.kubeconfig import build_config
 
-        server = master or os.environ.get("AGAC_API_SERVER", "")
-        if not server:
-            raise click.UsageError("--api http requires --master or $AGAC_API_SERVER")
-        kube_client = RestKubeClient(server)
+            kc = resolve_kubeconfig(kubeconfig)
+            if kc:
+                logger.info("Using kubeconfig: %s", kc)
+            else:
+                logger.info("Using in-cluster config")
+            kube_client = K8sKubeClient(build_config(master, kc))
+        else:
+            from .kube.rest import RestKubeClient
+
+            server = master or os.environ.get("AGAC_API_SERVER", "")
+            if not server:
+                raise click.UsageError("--api http requires --master or $AGAC_API_SERVER")
+            kube_client = RestKubeClient(server)
         try:
             from .cloudprovider.aws.client import boto3_cloud_factory
 
@@ -104,9 +115,6 @@ def controller(workers, cluster_name, kubeconfig, master, api, metrics_port, lea
 
             logger.warning("boto3 unavailable; falling back to the in-memory AWS fake")
             cloud_factory = FakeCloudFactory()
-        kc = resolve_kubeconfig(kubeconfig)
-        if kc:
-            logger.info("Using kubeconfig: %s", kc)
 
     if metrics_port:
         start_metrics_server(metrics_port)

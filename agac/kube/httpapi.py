@@ -93,9 +93,78 @@ class _Handler(BaseHTTPRequestHandler):
         query = {k: v[0] for k, v in parse_qs(parsed.query).items()}
         return parts, query
 
+    # -- k8s-compatible routes ---------------------------------------------
+    def _k8s_route(self, parts, query, method: str) -> bool:
+        """Serve real Kubernetes REST paths (``/api/v1/namespaces/...``,
+        ``/apis/<group>/<version>/...``, ``?watch=true``) so a stock
+        client-go-shaped client can talk to this server.  Returns True if
+        the path was a k8s route (handled), False to fall through."""
+        from . import k8swire
+
+        resolved = k8swire.resolve_path(parts)
+        if resolved is None:
+            return False
+        gvr, namespace, name, subresource = resolved
+        try:
+            if method == "GET" and name is None:
+                if query.get("watch") in ("true", "1"):
+                    rv = query.get("resourceVersion")
+                    self._serve_watch_stream(
+                        gvr.kind, namespace, int(rv) if rv else None, k8s_style=True
+                    )
+                    return True
+                items, rv = self.store.list(gvr.kind, namespace)
+                self._json(
+                    200,
+                    {
+                        "kind": f"{gvr.kind}List",
+                        "apiVersion": gvr.api_version,
+                        "metadata": {"resourceVersion": str(rv)},
+                        "items": [self._obj_with_kind(o) for o in items],
+                    },
+                )
+            elif method == "GET":
+                obj = self.store.get(gvr.kind, namespace or "", name)
+                self._json(200, self._obj_with_kind(obj))
+            elif method == "POST" and name is None:
+                cls = class_for_kind(gvr.kind)
+                obj = from_dict(cls, self._read_body())
+                if namespace:
+                    obj.metadata.namespace = namespace
+                created = self.store.create(obj)
+                self._json(201, self._obj_with_kind(created))
+            elif method == "PUT" and name is not None:
+                cls = class_for_kind(gvr.kind)
+                obj = from_dict(cls, self._read_body())
+                obj.metadata.namespace = namespace or obj.metadata.namespace
+                obj.metadata.name = name
+                if subresource == "status":
+                    updated = self.store.update_status(obj)
+                elif subresource is None:
+                    updated = self.store.update(obj)
+                else:
+                    self._json(404, k8swire.status_for_error(NotFoundError("unknown subresource")))
+                    return True
+                self._json(200, self._obj_with_kind(updated))
+            elif method == "DELETE" and name is not None:
+                self.store.delete(gvr.kind, namespace or "", name)
+                self._json(
+                    200,
+                    {"kind": "Status", "apiVersion": "v1", "status": "Success"},
+                )
+            else:
+                self._json(405, k8swire.status_for_error(APIError("method not allowed", 405)))
+        except APIError as e:
+            self._json(e.code, k8swire.status_for_error(e))
+        except BrokenPipeError:
+            pass
+        return True
+
     # -- verbs -------------------------------------------------------------
     def do_GET(self):  # noqa: N802
         parts, query = self._route()
+        if self._k8s_route(parts, query, "GET"):
+            return
         try:
             if parts == ["healthz"]:
                 self._json(200, {"status": "ok"})
@@ -118,7 +187,9 @@ class _Handler(BaseHTTPRequestHandler):
             pass
 
     def do_POST(self):  # noqa: N802
-        parts, _ = self._route()
+        parts, query = self._route()
+        if self._k8s_route(parts, query, "POST"):
+            return
         try:
             if len(parts) == 2 and parts[0] == "apis":
                 cls = class_for_kind(parts[1])
@@ -131,7 +202,9 @@ class _Handler(BaseHTTPRequestHandler):
             self._api_error(e)
 
     def do_PUT(self):  # noqa: N802
-        parts, _ = self._route()
+        parts, query = self._route()
+        if self._k8s_route(parts, query, "PUT"):
+            return
         try:
             if len(parts) in (4, 5) and parts[0] == "apis":
                 kind, ns, name = parts[1], parts[2], parts[3]
@@ -150,7 +223,9 @@ class _Handler(BaseHTTPRequestHandler):
             self._api_error(e)
 
     def do_DELETE(self):  # noqa: N802
-        parts, _ = self._route()
+        parts, query = self._route()
+        if self._k8s_route(parts, query, "DELETE"):
+            return
         try:
             if len(parts) == 4 and parts[0] == "apis":
                 self.store.delete(parts[1], parts[2], parts[3])
@@ -162,11 +237,17 @@ class _Handler(BaseHTTPRequestHandler):
 
     # -- watch streaming ----------------------------------------------------
     def _serve_watch(self, kind: str, query: dict):
-        namespace = query.get("namespace")
         rv = query.get("resourceVersion")
-        watch = self.store.watch(
-            kind, namespace, int(rv) if rv is not None else None
+        self._serve_watch_stream(
+            kind, query.get("namespace"), int(rv) if rv is not None else None,
+            k8s_style=False,
         )
+
+    def _serve_watch_stream(self, kind, namespace, rv, k8s_style: bool):
+        """Chunked ndjson event stream.  k8s_style frames events as the real
+        apiserver does ({"type", "object"}); the native scheme adds a
+        top-level resourceVersion."""
+        watch = self.store.watch(kind, namespace, rv)
         self.send_response(200)
         self.send_header("Content-Type", "application/json")
         self.send_header("Transfer-Encoding", "chunked")
@@ -178,14 +259,13 @@ class _Handler(BaseHTTPRequestHandler):
                     # heartbeat keeps half-open connections detectable
                     self._write_chunk(b"")
                     continue
-                line = json.dumps(
-                    {
-                        "type": event.type,
-                        "object": self._obj_with_kind(event.obj),
-                        "resourceVersion": event.resource_version,
-                    }
-                ).encode() + b"\n"
-                self._write_chunk(line)
+                payload = {
+                    "type": event.type,
+                    "object": self._obj_with_kind(event.obj),
+                }
+                if not k8s_style:
+                    payload["resourceVersion"] = event.resource_version
+                self._write_chunk(json.dumps(payload).encode() + b"\n")
         except (BrokenPipeError, ConnectionResetError, OSError):
             pass
         finally:

@@ -1,0 +1,290 @@
+"""Kubernetes wire-format compatibility: path registry, kubeconfig parsing,
+the API server's k8s-style routes (raw HTTP assertions on real apiserver
+shapes), and the K8sKubeClient + full manager over those routes."""
+
+import base64
+import json
+import threading
+import time
+import urllib.request
+
+import pytest
+
+from agac.apis import core as corev1
+from agac.apis.meta import ObjectMeta
+from agac.cloudprovider.aws.client import FakeCloudFactory
+from agac.cloudprovider.fake import FakeAWSBackend
+from agac.kube import k8swire
+from agac.kube.httpapi import APIServer
+from agac.kube.k8s import K8sKubeClient
+from agac.kube.kubeconfig import RestConfig, build_config, load_kubeconfig
+from agac.kube.store import APIStore, ConflictError, NotFoundError
+from agac.manager import ControllerConfig, Manager
+
+MANAGED = "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-managed"
+LB_TYPE = "service.beta.kubernetes.io/aws-load-balancer-type"
+
+
+def wait_until(pred, timeout=15.0):
+    deadline = time.monotonic() + timeout
+    while time.monotonic() < deadline:
+        try:
+            if pred():
+                return True
+        except Exception:
+            pass
+        time.sleep(0.02)
+    return False
+
+
+class TestPathRegistry:
+    def test_core_paths(self):
+        gvr = k8swire.gvr_for_kind("Service")
+        assert gvr.path() == "/api/v1/services"
+        assert gvr.path("default") == "/api/v1/namespaces/default/services"
+        assert gvr.path("default", "web") == "/api/v1/namespaces/default/services/web"
+        assert (
+            gvr.path("default", "web", "status")
+            == "/api/v1/namespaces/default/services/web/status"
+        )
+
+    def test_group_paths(self):
+        gvr = k8swire.gvr_for_kind("EndpointGroupBinding")
+        assert (
+            gvr.path("ns1", "b")
+            == "/apis/operator.h3poteto.dev/v1alpha1/namespaces/ns1/endpointgroupbindings/b"
+        )
+        assert gvr.api_version == "operator.h3poteto.dev/v1alpha1"
+        assert k8swire.gvr_for_kind("Ingress").path("d") == (
+            "/apis/networking.k8s.io/v1/namespaces/d/ingresses"
+        )
+        assert k8swire.gvr_for_kind("Lease").path("kube-system", "x") == (
+            "/apis/coordination.k8s.io/v1/namespaces/kube-system/leases/x"
+        )
+
+    def test_resolve_roundtrip(self):
+        for kind in ("Service", "Ingress", "Lease", "EndpointGroupBinding", "Event"):
+            gvr = k8swire.gvr_for_kind(kind)
+            parts = [p for p in gvr.path("ns", "n").split("/") if p]
+            resolved = k8swire.resolve_path(parts)
+            assert resolved is not None
+            assert resolved[0].kind == kind
+            assert resolved[1] == "ns" and resolved[2] == "n"
+
+    def test_resolve_rejects_unknown(self):
+        assert k8swire.resolve_path(["apis", "Service"]) is None  # native scheme
+        assert k8swire.resolve_path(["api", "v1", "pods"]) is None
+        assert k8swire.resolve_path(["healthz"]) is None
+
+
+class TestKubeconfig:
+    def test_token_and_inline_certs(self, tmp_path):
+        ca = base64.b64encode(b"CA PEM").decode()
+        config = {
+            "apiVersion": "v1",
+            "kind": "Config",
+            "current-context": "prod",
+            "contexts": [
+                {"name": "prod", "context": {"cluster": "c1", "user": "u1"}}
+            ],
+            "clusters": [
+                {
+                    "name": "c1",
+                    "cluster": {
+                        "server": "https://k8s.example.com:6443/",
+                        "certificate-authority-data": ca,
+                    },
+                }
+            ],
+            "users": [{"name": "u1", "user": {"token": "sekrit"}}],
+        }
+        import yaml
+
+        path = tmp_path / "kubeconfig"
+        path.write_text(yaml.safe_dump(config))
+        rest = load_kubeconfig(str(path))
+        assert rest.host == "https://k8s.example.com:6443"
+        assert rest.token == "sekrit"
+        assert rest.ca_cert is not None
+        assert open(rest.ca_cert, "rb").read() == b"CA PEM"
+        assert rest.verify == rest.ca_cert
+
+    def test_master_url_overrides(self, tmp_path):
+        import yaml
+
+        path = tmp_path / "kubeconfig"
+        path.write_text(
+            yaml.safe_dump(
+                {
+                    "current-context": "c",
+                    "contexts": [{"name": "c", "context": {"cluster": "x", "user": "y"}}],
+                    "clusters": [{"name": "x", "cluster": {"server": "https://a"}}],
+                    "users": [{"name": "y", "user": {}}],
+                }
+            )
+        )
+        rest = build_config(master_url="https://override:6443", kubeconfig=str(path))
+        assert rest.host == "https://override:6443"
+
+    def test_missing_context_errors(self, tmp_path):
+        path = tmp_path / "kubeconfig"
+        path.write_text("{}")
+        with pytest.raises(ValueError):
+            load_kubeconfig(str(path))
+
+
+@pytest.fixture
+def api():
+    server = APIServer(APIStore())
+    server.start()
+    yield server
+    server.shutdown()
+
+
+def raw(api, method, path, body=None):
+    req = urllib.request.Request(
+        api.url + path,
+        data=json.dumps(body).encode() if body is not None else None,
+        headers={"Content-Type": "application/json"},
+        method=method,
+    )
+    try:
+        with urllib.request.urlopen(req, timeout=5) as resp:
+            return resp.status, json.loads(resp.read())
+    except urllib.error.HTTPError as e:
+        return e.code, json.loads(e.read())
+
+
+import urllib.error  # noqa: E402
+
+
+class TestK8sRoutesRaw:
+    def test_create_and_get_service_k8s_shapes(self, api):
+        code, created = raw(
+            api,
+            "POST",
+            "/api/v1/namespaces/default/services",
+            {
+                "apiVersion": "v1",
+                "kind": "Service",
+                "metadata": {"name": "web"},
+                "spec": {"type": "LoadBalancer", "ports": [{"port": 80, "protocol": "TCP"}]},
+            },
+        )
+        assert code == 201
+        assert created["kind"] == "Service" and created["apiVersion"] == "v1"
+        assert created["metadata"]["resourceVersion"]
+
+        code, got = raw(api, "GET", "/api/v1/namespaces/default/services/web")
+        assert code == 200
+        assert got["spec"]["ports"][0]["port"] == 80
+
+        code, lst = raw(api, "GET", "/api/v1/namespaces/default/services")
+        assert code == 200
+        assert lst["kind"] == "ServiceList"
+        assert lst["metadata"]["resourceVersion"]
+        assert len(lst["items"]) == 1
+
+    def test_not_found_is_k8s_status_object(self, api):
+        code, status = raw(api, "GET", "/api/v1/namespaces/default/services/ghost")
+        assert code == 404
+        assert status["kind"] == "Status"
+        assert status["status"] == "Failure"
+        assert status["reason"] == "NotFound"
+
+    def test_delete_returns_status_success(self, api):
+        raw(
+            api, "POST", "/api/v1/namespaces/default/services",
+            {"metadata": {"name": "web"}, "spec": {}},
+        )
+        code, status = raw(api, "DELETE", "/api/v1/namespaces/default/services/web")
+        assert code == 200
+        assert status["kind"] == "Status" and status["status"] == "Success"
+
+    def test_crd_route(self, api):
+        code, created = raw(
+            api,
+            "POST",
+            "/apis/operator.h3poteto.dev/v1alpha1/namespaces/default/endpointgroupbindings",
+            {
+                "metadata": {"name": "b"},
+                "spec": {"endpointGroupArn": "arn:x", "clientIPPreservation": True},
+            },
+        )
+        assert code == 201
+        assert created["apiVersion"] == "operator.h3poteto.dev/v1alpha1"
+        assert created["spec"]["clientIPPreservation"] is True
+
+
+class TestK8sClient:
+    def client(self, api):
+        return K8sKubeClient(RestConfig(host=api.url))
+
+    def test_crud_roundtrip(self, api):
+        client = self.client(api)
+        client.create(
+            corev1.Service(
+                metadata=ObjectMeta(name="web", namespace="default"),
+                spec=corev1.ServiceSpec(type="LoadBalancer"),
+            )
+        )
+        got = client.get("Service", "default", "web")
+        assert got.spec.type == "LoadBalancer"
+        items, rv = client.list("Service", "default")
+        assert len(items) == 1 and rv > 0
+        got.metadata.annotations["k"] = "v"
+        client.update(got)
+        with pytest.raises(ConflictError):
+            client.update(got)  # stale rv
+        client.delete("Service", "default", "web")
+        with pytest.raises(NotFoundError):
+            client.get("Service", "default", "web")
+
+    def test_watch_k8s_framing(self, api):
+        client = self.client(api)
+        _, rv = client.list("Service")
+        watch = client.watch("Service", resource_version=rv)
+        try:
+            client.create(
+                corev1.Service(metadata=ObjectMeta(name="w", namespace="default"))
+            )
+            event = watch.get(timeout=10.0)
+            assert event is not None and event.type == "ADDED"
+            assert event.obj.metadata.name == "w"
+            assert event.resource_version > 0  # parsed from metadata
+        finally:
+            watch.stop()
+
+    def test_manager_reconciles_over_k8s_wire(self, api):
+        client = self.client(api)
+        backend = FakeAWSBackend()
+        stop = threading.Event()
+        manager = Manager()
+        manager.run(
+            client, ControllerConfig(), FakeCloudFactory(backend), stop,
+            resync_period=1.0, block=False,
+        )
+        try:
+            assert manager.wait_until_ready()
+            lb = backend.elbv2.create_load_balancer("wired", region="us-east-1")
+            client.create(
+                corev1.Service(
+                    metadata=ObjectMeta(
+                        name="wired",
+                        namespace="default",
+                        annotations={LB_TYPE: "nlb", MANAGED: "true"},
+                    ),
+                    spec=corev1.ServiceSpec(
+                        type="LoadBalancer",
+                        ports=[corev1.ServicePort(port=80, protocol="TCP")],
+                    ),
+                    status=corev1.ServiceStatus(
+                        load_balancer=corev1.LoadBalancerStatus(
+                            ingress=[corev1.LoadBalancerIngress(hostname=lb.dns_name)]
+                        )
+                    ),
+                )
+            )
+            assert wait_until(lambda: len(backend.ga.list_accelerators()[0]) == 1)
+        finally:
+            stop.set()
