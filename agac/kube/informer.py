@@ -164,6 +164,8 @@ class Informer:
             while not stop.is_set():
                 event = watch.get(timeout=0.2)
                 if event is None:
+                    if getattr(watch, "closed", False):
+                        return  # watch severed → relist
                     continue
                 self._handle_event(event)
         except GoneError:

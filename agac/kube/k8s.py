@@ -36,6 +36,7 @@ class _K8sWatch:
         self._cls = class_for_kind(kind)
         self._queue: "queue.Queue[Optional[WatchEvent]]" = queue.Queue()
         self._stopped = False
+        self.closed = False
         self._thread = threading.Thread(target=self._pump, daemon=True)
         self._thread.start()
 
@@ -62,10 +63,15 @@ class _K8sWatch:
             self._queue.put(None)
 
     def get(self, timeout: Optional[float] = None) -> Optional[WatchEvent]:
+        if self.closed:
+            return None
         try:
-            return self._queue.get(timeout=timeout)
+            item = self._queue.get(timeout=timeout)
         except queue.Empty:
             return None
+        if item is None:
+            self.closed = True
+        return item
 
     def stop(self):
         self._stopped = True

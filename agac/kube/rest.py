@@ -62,6 +62,7 @@ class _RestWatch:
         self._response = response
         self._queue: "queue.Queue[Optional[WatchEvent]]" = queue.Queue()
         self._stopped = False
+        self.closed = False
         self._thread = threading.Thread(target=self._pump, daemon=True)
         self._thread.start()
 
@@ -88,12 +89,14 @@ class _RestWatch:
             self._queue.put(None)
 
     def get(self, timeout: Optional[float] = None) -> Optional[WatchEvent]:
+        if self.closed:
+            return None
         try:
             item = self._queue.get(timeout=timeout)
         except queue.Empty:
             return None
-        if item is None and self._stopped:
-            return None
+        if item is None:
+            self.closed = True
         return item
 
     def stop(self):

@@ -89,6 +89,7 @@ class _Watch:
         self._kind = kind
         self._namespace = namespace
         self._stopped = False
+        self.closed = False  # set once the end-of-stream sentinel is seen
 
     def _accepts(self, kind: str, namespace: str) -> bool:
         if kind != self._kind:
@@ -114,10 +115,17 @@ class _Watch:
         return item
 
     def get(self, timeout: Optional[float] = None) -> Optional[WatchEvent]:
+        """None means timeout OR end-of-stream; check ``closed`` to tell
+        them apart (a closed watch makes the informer relist)."""
+        if self.closed:
+            return None
         try:
-            return self._queue.get(timeout=timeout)
+            item = self._queue.get(timeout=timeout)
         except queue.Empty:
             return None
+        if item is None:
+            self.closed = True
+        return item
 
 
 class APIStore:
