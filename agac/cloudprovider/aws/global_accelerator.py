@@ -507,6 +507,33 @@ class GlobalAcceleratorMixin:
         )
         logger.info("Endpoint weight is updated: %s", endpoint_id)
 
+    def sync_endpoint_weights(
+        self,
+        endpoint_group: t.EndpointGroup,
+        endpoint_ids,
+        weight: Optional[int],
+    ):
+        """Set the weight of every id in ``endpoint_ids`` in one
+        describe + one update (the reference loops per-endpoint
+        UpdateEndpointGroup calls, egb/reconcile.go:199-206 — K×2 API
+        round-trips instead of 2)."""
+        metrics.observe_aws_call("globalaccelerator", "DescribeEndpointGroup")
+        current = self.ga.describe_endpoint_group(endpoint_group.endpoint_group_arn)
+        target = set(endpoint_ids)
+        configs = [
+            t.EndpointConfiguration(
+                endpoint_id=d.endpoint_id,
+                weight=weight if d.endpoint_id in target else d.weight,
+                client_ip_preservation_enabled=d.client_ip_preservation_enabled,
+            )
+            for d in current.endpoint_descriptions
+        ]
+        metrics.observe_aws_call("globalaccelerator", "UpdateEndpointGroup")
+        self.ga.update_endpoint_group(
+            endpoint_group.endpoint_group_arn, endpoint_configurations=configs
+        )
+        logger.info("Endpoint weights are synced for %d endpoints", len(target))
+
     def describe_endpoint_group(self, endpoint_group_arn: str) -> t.EndpointGroup:
         metrics.observe_aws_call("globalaccelerator", "DescribeEndpointGroup")
         return self.ga.describe_endpoint_group(endpoint_group_arn)

@@ -255,10 +255,12 @@ class EndpointGroupBindingController:
             if added_id is not None:
                 results.append(added_id)
 
-        for endpoint_id, (_, region) in arns.items():
-            regional_cloud = self.cloud_factory(region)
-            regional_cloud.update_endpoint_weight(
-                endpoint_group, endpoint_id, binding.spec.weight
+        if arns:
+            # batched weight sync: one describe + one update for all ids
+            # (regions of the LBs are irrelevant — GA endpoint groups are
+            # managed through the us-west-2-homed client)
+            cloud.sync_endpoint_weights(
+                endpoint_group, list(arns), binding.spec.weight
             )
 
         copied = deep_copy(binding)

@@ -10,6 +10,8 @@ from .store import APIStore, ConflictError, NotFoundError, AlreadyExistsError, i
 from .client import KubeClient, InMemoryKubeClient
 from .workqueue import RateLimitingQueue, ItemExponentialFailureRateLimiter
 from .informer import SharedInformerFactory, Informer, Lister, wait_for_cache_sync
+from .leaderelection import LeaderElector, LeaderElectionConfig
+from .admission import AdmissionDeniedError, http_admission, local_admission
 
 __all__ = [
     "APIStore",
@@ -25,4 +27,9 @@ __all__ = [
     "Informer",
     "Lister",
     "wait_for_cache_sync",
+    "LeaderElector",
+    "LeaderElectionConfig",
+    "AdmissionDeniedError",
+    "http_admission",
+    "local_admission",
 ]
