@@ -119,10 +119,13 @@ _ERROR_REASONS = {
 def status_for_error(e: APIError) -> dict:
     """k8s metav1.Status failure object."""
     from .admission import AdmissionDeniedError
+    from .validation import ValidationError
 
     reason = _ERROR_REASONS.get(type(e))
     if reason is None and isinstance(e, AdmissionDeniedError):
         reason = "Forbidden"
+    if reason is None and isinstance(e, ValidationError):
+        reason = "Invalid"
     return {
         "kind": "Status",
         "apiVersion": "v1",
@@ -138,6 +141,8 @@ def error_for_status(status: dict, http_code: int) -> APIError:
 
     reason = status.get("reason", "")
     message = status.get("message", "")
+    from .validation import ValidationError
+
     mapping = {
         "NotFound": NotFoundError,
         "AlreadyExists": AlreadyExistsError,
@@ -145,6 +150,7 @@ def error_for_status(status: dict, http_code: int) -> APIError:
         "Expired": GoneError,
         "Gone": GoneError,
         "Forbidden": AdmissionDeniedError,
+        "Invalid": ValidationError,
     }
     cls = mapping.get(reason)
     if cls is not None:

@@ -175,6 +175,9 @@ class APIStore:
     # -- CRUD --------------------------------------------------------------
     def create(self, obj):
         kind = type(obj).kind
+        from .validation import validate_object
+
+        validate_object(obj)
         self._admit(kind, "CREATE", None, obj)
         with self._lock:
             key = (obj.metadata.namespace, obj.metadata.name)
@@ -229,6 +232,9 @@ class APIStore:
         """Update spec+metadata.  Bumps generation if spec changed; removes the
         object if it has a deletionTimestamp and finalizers became empty."""
         kind = type(obj).kind
+        from .validation import validate_object
+
+        validate_object(obj)
         try:
             current = self.get(kind, obj.metadata.namespace, obj.metadata.name)
         except NotFoundError:
