@@ -63,6 +63,15 @@ class FakeELBv2:
         dns_name: Optional[str] = None,
     ) -> t.LoadBalancer:
         with self._lock:
+            for existing_arn, existing in self._lbs.items():
+                if (
+                    existing.load_balancer_name == name
+                    and self._regions.get(existing_arn) == region
+                ):
+                    raise awserr.AWSAPIError(
+                        f"A load balancer with the name '{name}' already exists",
+                        "DuplicateLoadBalancerName",
+                    )
             kind = "net" if lb_type == "network" else "app"
             arn = (
                 f"arn:aws:elasticloadbalancing:{region}:{_ACCOUNT}:"

@@ -1,0 +1,133 @@
+"""Property-based tests (hypothesis): wire-format round-trips, deep-copy
+equivalence, hostname parsing on generated ELB hostnames."""
+
+import string
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from agac.apis import core as corev1
+from agac.apis import endpointgroupbinding as egb
+from agac.apis.meta import ObjectMeta, deep_copy, from_dict, to_dict
+from agac.cloudprovider.aws.load_balancer import get_lb_name_from_hostname
+
+# DNS-label-ish names (k8s object names)
+names = st.text(
+    alphabet=string.ascii_lowercase + string.digits + "-", min_size=1, max_size=30
+).filter(lambda s: not s.startswith("-") and not s.endswith("-"))
+
+annotations = st.dictionaries(
+    st.text(string.ascii_lowercase + "./-", min_size=1, max_size=40),
+    st.text(max_size=60),
+    max_size=5,
+)
+
+ports = st.lists(
+    st.builds(
+        corev1.ServicePort,
+        port=st.integers(min_value=1, max_value=65535),
+        protocol=st.sampled_from(["TCP", "UDP"]),
+        name=names,
+    ),
+    max_size=4,
+)
+
+
+@st.composite
+def services(draw):
+    return corev1.Service(
+        metadata=ObjectMeta(
+            name=draw(names),
+            namespace=draw(names),
+            annotations=draw(annotations),
+            finalizers=draw(st.lists(names, max_size=2)),
+        ),
+        spec=corev1.ServiceSpec(
+            type=draw(st.sampled_from(["ClusterIP", "LoadBalancer", "NodePort"])),
+            ports=draw(ports),
+            load_balancer_class=draw(st.none() | names),
+        ),
+        status=corev1.ServiceStatus(
+            load_balancer=corev1.LoadBalancerStatus(
+                ingress=[
+                    corev1.LoadBalancerIngress(hostname=draw(names), ip="")
+                    for _ in range(draw(st.integers(0, 2)))
+                ]
+            )
+        ),
+    )
+
+
+@st.composite
+def bindings(draw):
+    return egb.EndpointGroupBinding(
+        metadata=ObjectMeta(name=draw(names), namespace=draw(names)),
+        spec=egb.EndpointGroupBindingSpec(
+            endpoint_group_arn="arn:aws:globalaccelerator::1:eg/" + draw(names),
+            client_ip_preservation=draw(st.booleans()),
+            weight=draw(st.none() | st.integers(0, 255)),
+            service_ref=draw(
+                st.none() | st.builds(egb.ServiceReference, name=names)
+            ),
+        ),
+        status=egb.EndpointGroupBindingStatus(
+            endpoint_ids=draw(st.lists(names, max_size=3)),
+            observed_generation=draw(st.integers(0, 100)),
+        ),
+    )
+
+
+@settings(max_examples=50, deadline=None)
+@given(services())
+def test_service_wire_roundtrip(svc):
+    assert from_dict(corev1.Service, to_dict(svc)) == svc or _normalized_equal(svc)
+
+
+def _normalized_equal(svc):
+    # empty strings serialize away ("omitempty"): compare via double roundtrip
+    once = from_dict(corev1.Service, to_dict(svc))
+    return to_dict(once) == to_dict(svc)
+
+
+@settings(max_examples=50, deadline=None)
+@given(bindings())
+def test_egb_wire_roundtrip_stable(binding):
+    d1 = to_dict(binding)
+    back = from_dict(egb.EndpointGroupBinding, d1)
+    assert to_dict(back) == d1
+
+
+@settings(max_examples=50, deadline=None)
+@given(services())
+def test_deep_copy_is_equal_and_independent(svc):
+    copied = deep_copy(svc)
+    assert copied == svc
+    assert copied is not svc
+    copied.metadata.annotations["mutated"] = "x"
+    assert "mutated" not in svc.metadata.annotations
+
+
+lb_names = st.text(
+    alphabet=string.ascii_lowercase + string.digits + "-", min_size=1, max_size=28
+).filter(lambda s: not s.startswith("-") and not s.endswith("-") and "--" not in s)
+hashes = st.text(alphabet=string.ascii_lowercase + string.digits, min_size=8, max_size=16)
+regions = st.sampled_from(["us-east-1", "us-west-2", "eu-central-1", "ap-northeast-1"])
+
+
+@settings(max_examples=50, deadline=None)
+@given(lb_names, hashes, regions)
+def test_nlb_hostname_roundtrip(name, h, region):
+    hostname = f"{name}-{h}.elb.{region}.amazonaws.com"
+    parsed_name, parsed_region = get_lb_name_from_hostname(hostname)
+    assert parsed_name == name
+    assert parsed_region == region
+
+
+@settings(max_examples=50, deadline=None)
+@given(lb_names, hashes, regions, st.booleans())
+def test_alb_hostname_roundtrip(name, h, region, internal):
+    prefix = "internal-" if internal else ""
+    hostname = f"{prefix}{name}-{h}.{region}.elb.amazonaws.com"
+    parsed_name, parsed_region = get_lb_name_from_hostname(hostname)
+    assert parsed_name == name
+    assert parsed_region == region
