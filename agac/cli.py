@@ -69,9 +69,10 @@ def resolve_kubeconfig(kubeconfig: str) -> str:
 @click.option("--kubeconfig", default="", help="Path to a kubeconfig. Only required if out-of-cluster.")
 @click.option("--master", default="", help="The address of the Kubernetes API server. Overrides any value in kubeconfig.")
 @click.option("--api", type=click.Choice(["memory", "http", "k8s"]), default="memory", show_default=True, help="Kube API backend: embedded in-memory store, an agac HTTP API server, or a real Kubernetes API server (kubeconfig/in-cluster auth).")
+@click.option("--cloud", type=click.Choice(["auto", "aws", "fake"]), default="auto", show_default=True, help="Cloud backend: auto (aws for http/k8s APIs, fake for memory), aws (boto3, required), fake (in-memory).")
 @click.option("--metrics-port", default=0, help="Serve Prometheus metrics on this port (0 = disabled).")
 @click.option("--leader-elect/--no-leader-elect", default=True, show_default=True)
-def controller(workers, cluster_name, kubeconfig, master, api, metrics_port, leader_elect):
+def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_port, leader_elect):
     """Start controller."""
     from .controller.endpointgroupbinding import EndpointGroupBindingConfig
     from .controller.globalaccelerator import GlobalAcceleratorConfig
@@ -82,39 +83,47 @@ def controller(workers, cluster_name, kubeconfig, master, api, metrics_port, lea
     from .signals import setup_signal_handler
 
     if api == "memory":
-        from .cloudprovider.aws.client import FakeCloudFactory
         from .kube.client import InMemoryKubeClient
 
         kube_client = InMemoryKubeClient()
-        cloud_factory = FakeCloudFactory()
-        logger.info("Using embedded in-memory API store + AWS fake")
-    else:
-        if api == "k8s":
-            from .kube.k8s import K8sKubeClient
-            from .kube.kubeconfig import build_config
+        logger.info("Using embedded in-memory API store")
+    elif api == "k8s":
+        from .kube.k8s import K8sKubeClient
+        from .kube.kubeconfig import build_config
 
-            kc = resolve_kubeconfig(kubeconfig)
-            if kc:
-                logger.info("Using kubeconfig: %s", kc)
-            else:
-                logger.info("Using in-cluster config")
-            kube_client = K8sKubeClient(build_config(master, kc))
+        kc = resolve_kubeconfig(kubeconfig)
+        if kc:
+            logger.info("Using kubeconfig: %s", kc)
         else:
-            from .kube.rest import RestKubeClient
+            logger.info("Using in-cluster config")
+        kube_client = K8sKubeClient(build_config(master, kc))
+    else:
+        from .kube.rest import RestKubeClient
 
-            server = master or os.environ.get("AGAC_API_SERVER", "")
-            if not server:
-                raise click.UsageError("--api http requires --master or $AGAC_API_SERVER")
-            kube_client = RestKubeClient(server)
+        server = master or os.environ.get("AGAC_API_SERVER", "")
+        if not server:
+            raise click.UsageError("--api http requires --master or $AGAC_API_SERVER")
+        kube_client = RestKubeClient(server)
+
+    # Cloud backend selection is explicit: a controller that silently
+    # mutates a throwaway fake while looking healthy would be dangerous,
+    # so "auto" only picks the fake for the embedded API.
+    if cloud == "auto":
+        cloud = "fake" if api == "memory" else "aws"
+    if cloud == "aws":
+        from .cloudprovider.aws.client import boto3_cloud_factory
+
         try:
-            from .cloudprovider.aws.client import boto3_cloud_factory
-
             cloud_factory = boto3_cloud_factory()
-        except RuntimeError:
-            from .cloudprovider.aws.client import FakeCloudFactory
+        except RuntimeError as e:
+            raise click.UsageError(
+                f"{e} (use --cloud fake for a demo without AWS credentials)"
+            ) from e
+    else:
+        from .cloudprovider.aws.client import FakeCloudFactory
 
-            logger.warning("boto3 unavailable; falling back to the in-memory AWS fake")
-            cloud_factory = FakeCloudFactory()
+        cloud_factory = FakeCloudFactory()
+        logger.info("Using the in-memory AWS fake")
 
     if metrics_port:
         start_metrics_server(metrics_port)

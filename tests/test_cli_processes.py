@@ -72,6 +72,7 @@ def test_apiserver_and_controller_processes():
                 "-v", "controller",
                 "--api", "http",
                 "--master", "http://127.0.0.1:18001",
+                "--cloud", "fake",
                 "--workers", "1",
             ]
         )
