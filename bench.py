@@ -35,9 +35,11 @@ import threading
 import time
 
 
-def build_stack(objects: int, workers: int):
+def build_stack(objects: int, workers: int, scenario: str = "ga"):
     from agac.apis import core as corev1
+    from agac.apis import endpointgroupbinding as egb
     from agac.apis.meta import ObjectMeta
+    from agac.cloudprovider.aws import types as awstypes
     from agac.cloudprovider.aws.client import FakeCloudFactory
     from agac.cloudprovider.fake import FakeAWSBackend
     from agac.controller.endpointgroupbinding import EndpointGroupBindingConfig
@@ -49,6 +51,7 @@ def build_stack(objects: int, workers: int):
     MANAGED = (
         "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-managed"
     )
+    HOSTNAME = "aws-global-accelerator-controller.h3poteto.dev/route53-hostname"
     LB_TYPE = "service.beta.kubernetes.io/aws-load-balancer-type"
     region = "us-east-1"
 
@@ -71,14 +74,26 @@ def build_stack(objects: int, workers: int):
     if not manager.wait_until_ready():
         raise RuntimeError("controllers did not become ready")
 
+    if scenario == "full":
+        backend.route53.create_hosted_zone("bench.example.com")
+        ext_acc = backend.ga.create_accelerator("external-bench")
+        ext_listener = backend.ga.create_listener(
+            ext_acc.accelerator_arn, [awstypes.PortRange(80, 80)], "TCP"
+        )
+        ext_group = backend.ga.create_endpoint_group(ext_listener.listener_arn, region)
+
     services = []
+    bindings = []
     for i in range(objects):
         lb = backend.elbv2.create_load_balancer(f"lb-{i}", region=region)
+        annotations = {LB_TYPE: "nlb", MANAGED: "true"}
+        if scenario == "full":
+            annotations[HOSTNAME] = f"svc-{i}.bench.example.com"
         svc = corev1.Service(
             metadata=ObjectMeta(
                 name=f"svc-{i}",
                 namespace="default",
-                annotations={LB_TYPE: "nlb", MANAGED: "true"},
+                annotations=annotations,
             ),
             spec=corev1.ServiceSpec(
                 type="LoadBalancer",
@@ -93,7 +108,20 @@ def build_stack(objects: int, workers: int):
         client.create(svc)
         services.append(svc.metadata.name)
 
-    return client, backend, services, stop
+    if scenario == "full":
+        for i in range(max(1, objects // 4)):
+            binding = egb.EndpointGroupBinding(
+                metadata=ObjectMeta(name=f"bind-{i}", namespace="default"),
+                spec=egb.EndpointGroupBindingSpec(
+                    endpoint_group_arn=ext_group.endpoint_group_arn,
+                    weight=100,
+                    service_ref=egb.ServiceReference(name=f"svc-{i}"),
+                ),
+            )
+            client.create(binding)
+            bindings.append(binding.metadata.name)
+
+    return client, backend, services, bindings, stop
 
 
 def converged(backend, owner_to_port: dict) -> bool:
@@ -120,17 +148,41 @@ def converged(backend, owner_to_port: dict) -> bool:
         return seen == len(owner_to_port)
 
 
-def run_step(client, backend, services, step_idx: int, timeout: float = 120.0):
-    """Mutate every service's port and wait for full convergence."""
+def bindings_converged(client, backend, bindings, weight: int) -> bool:
+    """Every binding's endpoint attached with the expected weight."""
+    for name in bindings:
+        binding = client.get("EndpointGroupBinding", "default", name)
+        if not binding.status.endpoint_ids:
+            return False
+        group = backend.ga.describe_endpoint_group(binding.spec.endpoint_group_arn)
+        weights = {d.endpoint_id: d.weight for d in group.endpoint_descriptions}
+        for endpoint_id in binding.status.endpoint_ids:
+            if weights.get(endpoint_id) != weight:
+                return False
+    return True
+
+
+def run_step(client, backend, services, step_idx: int, timeout: float = 120.0,
+             bindings=()):
+    """Mutate every service's port (and every binding's weight in the full
+    scenario) and wait for full convergence."""
     port = 8000 + (step_idx % 2)
+    weight = 100 + (step_idx % 2)
     owner_to_port = {}
     for name in services:
         svc = client.get("Service", "default", name)
         svc.spec.ports[0].port = port
         client.update(svc)
         owner_to_port[f"service/default/{name}"] = port
+    for name in bindings:
+        binding = client.get("EndpointGroupBinding", "default", name)
+        binding.spec.weight = weight
+        client.update(binding)
     deadline = time.monotonic() + timeout
-    while not converged(backend, owner_to_port):
+    while not (
+        converged(backend, owner_to_port)
+        and (not bindings or bindings_converged(client, backend, bindings, weight))
+    ):
         if time.monotonic() >= deadline:
             raise TimeoutError(f"step {step_idx} did not converge in {timeout}s")
         time.sleep(0.001)
@@ -146,6 +198,11 @@ def main():
     # setting under the GIL for this CPU-bound fake (measured: 1w=535/s,
     # 8w=234/s on this container)
     parser.add_argument("--workers", type=int, default=1)
+    parser.add_argument(
+        "--scenario", choices=["ga", "full"], default="ga",
+        help="ga: GlobalAccelerator service churn (the headline metric); "
+             "full: adds Route53 records + EndpointGroupBinding weight churn",
+    )
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -158,7 +215,9 @@ def main():
         # timing across the per-GPU ranks without touching the GPUs
         dist.init_process_group(backend="gloo")
 
-    client, backend, services, stop = build_stack(args.objects, args.workers)
+    client, backend, services, bindings, stop = build_stack(
+        args.objects, args.workers, args.scenario
+    )
 
     try:
         # initial creation converges during warmup setup
@@ -170,7 +229,7 @@ def main():
             time.sleep(0.001)
 
         for w in range(args.warmup):
-            run_step(client, backend, services, w)
+            run_step(client, backend, services, w, bindings=bindings)
 
         try:
             import torch
@@ -185,7 +244,7 @@ def main():
             torch.cuda.synchronize()
         start = time.monotonic()
         for k in range(args.steps):
-            run_step(client, backend, services, args.warmup + k)
+            run_step(client, backend, services, args.warmup + k, bindings=bindings)
         if cuda:
             torch.cuda.synchronize()
         elapsed = time.monotonic() - start
@@ -198,8 +257,9 @@ def main():
             elapsed = float(t.item())
 
         ms_per_step = elapsed / args.steps * 1000.0
-        # whole-job aggregate: every rank converged `objects` objects per step
-        value = args.objects * args.steps * world_size / elapsed
+        # whole-job aggregate: every rank converged its objects per step
+        objects_per_step = args.objects + len(bindings)
+        value = objects_per_step * args.steps * world_size / elapsed
 
         if rank == 0:
             print(
@@ -220,6 +280,8 @@ def main():
                         "config": {
                             "model": "k8s-controller reconcile loop (BASELINE.json: tier-mismatch, no ML model; proxy metric = reconcile latency event->converged)",
                             "objects_per_rank": args.objects,
+                            "scenario": args.scenario,
+                            "bindings_per_rank": len(bindings),
                             "workers_per_queue": args.workers,
                             "parallelism": f"independent controller stack per rank (x{world_size})",
                         },

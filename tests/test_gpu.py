@@ -27,7 +27,7 @@ def test_bench_entry_one_step():
     """bench.py's stack builds and converges one step on the box."""
     import bench
 
-    client, backend, services, stop = bench.build_stack(objects=4, workers=2)
+    client, backend, services, bindings, stop = bench.build_stack(objects=4, workers=2)
     try:
         bench.run_step(client, backend, services, 0, timeout=60.0)
     finally:
