@@ -121,9 +121,14 @@ def test_lost_lease_stops_leading():
     stop = threading.Event()
     t = run_in_thread(ea, stop)
     assert sa.wait(timeout=5.0)
-    # usurp the lease out-of-band (simulates apiserver-side takeover)
+    # usurp the lease out-of-band (simulates apiserver-side takeover); give
+    # the intruder a long duration so "a" cannot reacquire via expiry while
+    # its renew loop is still timing out
     lease = client.get("Lease", "kube-system", "agac-leader")
     lease.spec.holder_identity = "intruder"
+    lease.spec.lease_duration_seconds = 3600
+    from agac.kube.leaderelection import _fmt, _now
+    lease.spec.renew_time = _fmt(_now())
     client.update(lease)
     deadline = time.monotonic() + 5
     while ea.is_leader.is_set() and time.monotonic() < deadline:
