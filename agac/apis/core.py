@@ -21,7 +21,7 @@ EVENT_TYPE_WARNING = "Warning"
 # ---------------------------------------------------------------------------
 # Service
 # ---------------------------------------------------------------------------
-@dataclass
+@dataclass(slots=True)
 class ServicePort:
     name: str = ""
     protocol: str = "TCP"
@@ -30,7 +30,7 @@ class ServicePort:
     node_port: typing.Optional[int] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class ServiceSpec:
     type: str = "ClusterIP"
     ports: typing.List[ServicePort] = field(default_factory=list)
@@ -38,31 +38,31 @@ class ServiceSpec:
     selector: typing.Dict[str, str] = field(default_factory=dict)
 
 
-@dataclass
+@dataclass(slots=True)
 class PortStatus:
     port: int = 0
     protocol: str = "TCP"
     error: typing.Optional[str] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class LoadBalancerIngress:
     ip: str = ""
     hostname: str = ""
     ports: typing.List[PortStatus] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class LoadBalancerStatus:
     ingress: typing.List[LoadBalancerIngress] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class ServiceStatus:
     load_balancer: LoadBalancerStatus = field(default_factory=LoadBalancerStatus)
 
 
-@dataclass
+@dataclass(slots=True)
 class Service:
     kind: typing.ClassVar[str] = "Service"
     api_version: typing.ClassVar[str] = "v1"
@@ -75,75 +75,75 @@ class Service:
 # ---------------------------------------------------------------------------
 # Ingress (networking.k8s.io/v1)
 # ---------------------------------------------------------------------------
-@dataclass
+@dataclass(slots=True)
 class ServiceBackendPort:
     name: str = ""
     number: int = 0
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressServiceBackend:
     name: str = ""
     port: ServiceBackendPort = field(default_factory=ServiceBackendPort)
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressBackend:
     service: typing.Optional[IngressServiceBackend] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class HTTPIngressPath:
     path: str = ""
     path_type: str = "Prefix"
     backend: IngressBackend = field(default_factory=IngressBackend)
 
 
-@dataclass
+@dataclass(slots=True)
 class HTTPIngressRuleValue:
     paths: typing.List[HTTPIngressPath] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressRule:
     host: str = ""
     http: typing.Optional[HTTPIngressRuleValue] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressSpec:
     ingress_class_name: typing.Optional[str] = None
     default_backend: typing.Optional[IngressBackend] = None
     rules: typing.List[IngressRule] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressPortStatus:
     port: int = 0
     protocol: str = "TCP"
     error: typing.Optional[str] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressLoadBalancerIngress:
     ip: str = ""
     hostname: str = ""
     ports: typing.List[IngressPortStatus] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressLoadBalancerStatus:
     ingress: typing.List[IngressLoadBalancerIngress] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressStatus:
     load_balancer: IngressLoadBalancerStatus = field(
         default_factory=IngressLoadBalancerStatus
     )
 
 
-@dataclass
+@dataclass(slots=True)
 class Ingress:
     kind: typing.ClassVar[str] = "Ingress"
     api_version: typing.ClassVar[str] = "networking.k8s.io/v1"
@@ -156,7 +156,7 @@ class Ingress:
 # ---------------------------------------------------------------------------
 # Event (recorded by controllers, reference record.EventRecorder)
 # ---------------------------------------------------------------------------
-@dataclass
+@dataclass(slots=True)
 class ObjectReference:
     kind: str = ""
     namespace: str = ""
@@ -164,12 +164,12 @@ class ObjectReference:
     uid: str = ""
 
 
-@dataclass
+@dataclass(slots=True)
 class EventSource:
     component: str = ""
 
 
-@dataclass
+@dataclass(slots=True)
 class Event:
     kind: typing.ClassVar[str] = "Event"
     api_version: typing.ClassVar[str] = "v1"
@@ -188,7 +188,7 @@ class Event:
 # ---------------------------------------------------------------------------
 # Lease (coordination.k8s.io/v1, used by leader election)
 # ---------------------------------------------------------------------------
-@dataclass
+@dataclass(slots=True)
 class LeaseSpec:
     holder_identity: typing.Optional[str] = None
     lease_duration_seconds: typing.Optional[int] = None
@@ -197,7 +197,7 @@ class LeaseSpec:
     lease_transitions: int = 0
 
 
-@dataclass
+@dataclass(slots=True)
 class Lease:
     kind: typing.ClassVar[str] = "Lease"
     api_version: typing.ClassVar[str] = "coordination.k8s.io/v1"

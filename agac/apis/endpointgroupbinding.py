@@ -17,17 +17,17 @@ VERSION = "v1alpha1"
 FINALIZER = "operator.h3poteto.dev/endpointgroupbindings"
 
 
-@dataclass
+@dataclass(slots=True)
 class ServiceReference:
     name: str = ""
 
 
-@dataclass
+@dataclass(slots=True)
 class IngressReference:
     name: str = ""
 
 
-@dataclass
+@dataclass(slots=True)
 class EndpointGroupBindingSpec:
     _json_overrides: typing.ClassVar[dict] = {
         "client_ip_preservation": "clientIPPreservation",
@@ -40,7 +40,7 @@ class EndpointGroupBindingSpec:
     ingress_ref: typing.Optional[IngressReference] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class EndpointGroupBindingStatus:
     _keep_empty: typing.ClassVar[set] = {"endpoint_ids"}
 
@@ -48,7 +48,7 @@ class EndpointGroupBindingStatus:
     observed_generation: int = 0
 
 
-@dataclass
+@dataclass(slots=True)
 class EndpointGroupBinding:
     kind: typing.ClassVar[str] = "EndpointGroupBinding"
     api_version: typing.ClassVar[str] = f"{GROUP}/{VERSION}"

@@ -90,7 +90,7 @@ def from_dict(cls, data: dict):
     return cls(**kwargs)
 
 
-@dataclass
+@dataclass(slots=True)
 class ObjectMeta:
     """Reference: metav1.ObjectMeta (the subset the controllers use)."""
 
@@ -106,7 +106,7 @@ class ObjectMeta:
     finalizers: typing.List[str] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class TypeMeta:
     kind: str = ""
     api_version: str = ""

@@ -32,13 +32,13 @@ CHANGE_ACTION_DELETE = "DELETE"
 GLOBAL_ACCELERATOR_HOSTED_ZONE_ID = "Z2BJ6XQ5FK7U4H"
 
 
-@dataclass
+@dataclass(slots=True)
 class Tag:
     key: str
     value: str
 
 
-@dataclass
+@dataclass(slots=True)
 class Accelerator:
     accelerator_arn: str = ""
     name: str = ""
@@ -48,13 +48,13 @@ class Accelerator:
     ip_address_type: str = IP_ADDRESS_TYPE_DUAL_STACK
 
 
-@dataclass
+@dataclass(slots=True)
 class PortRange:
     from_port: int = 0
     to_port: int = 0
 
 
-@dataclass
+@dataclass(slots=True)
 class Listener:
     listener_arn: str = ""
     port_ranges: typing.List[PortRange] = field(default_factory=list)
@@ -62,7 +62,7 @@ class Listener:
     client_affinity: str = CLIENT_AFFINITY_NONE
 
 
-@dataclass
+@dataclass(slots=True)
 class EndpointDescription:
     endpoint_id: str = ""
     weight: typing.Optional[int] = None
@@ -70,21 +70,21 @@ class EndpointDescription:
     health_state: str = "HEALTHY"
 
 
-@dataclass
+@dataclass(slots=True)
 class EndpointGroup:
     endpoint_group_arn: str = ""
     endpoint_group_region: str = ""
     endpoint_descriptions: typing.List[EndpointDescription] = field(default_factory=list)
 
 
-@dataclass
+@dataclass(slots=True)
 class EndpointConfiguration:
     endpoint_id: str = ""
     weight: typing.Optional[int] = None
     client_ip_preservation_enabled: typing.Optional[bool] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class LoadBalancer:
     load_balancer_arn: str = ""
     load_balancer_name: str = ""
@@ -94,25 +94,25 @@ class LoadBalancer:
     scheme: str = "internet-facing"
 
 
-@dataclass
+@dataclass(slots=True)
 class HostedZone:
     id: str = ""
     name: str = ""  # always dot-terminated, e.g. "example.com."
 
 
-@dataclass
+@dataclass(slots=True)
 class AliasTarget:
     dns_name: str = ""
     evaluate_target_health: bool = True
     hosted_zone_id: str = ""
 
 
-@dataclass
+@dataclass(slots=True)
 class ResourceRecord:
     value: str = ""
 
 
-@dataclass
+@dataclass(slots=True)
 class ResourceRecordSet:
     name: str = ""  # dot-terminated, wildcards octal-escaped ("\\052")
     type: str = RR_TYPE_A
@@ -121,7 +121,7 @@ class ResourceRecordSet:
     alias_target: typing.Optional[AliasTarget] = None
 
 
-@dataclass
+@dataclass(slots=True)
 class Change:
     action: str
     record_set: ResourceRecordSet
