@@ -26,9 +26,9 @@ from .store import (
     WatchEvent,
 )
 
-logger = logging.getLogger(__name__)
-
 from .admission import AdmissionDeniedError
+
+logger = logging.getLogger(__name__)
 
 _REASON_TO_ERROR = {
     "NotFound": NotFoundError,
