@@ -195,12 +195,74 @@ def version():
 
 @cli.command("apiserver")
 @click.option("--port", default=8001, show_default=True, help="Listen port.")
-def apiserver(port):
+@click.option("--state-file", default="", help="Snapshot file: loaded on start if present, written on shutdown (checkpoint/resume).")
+def apiserver(port, state_file):
     """Serve the in-memory API store over HTTP (hermetic e2e backend)."""
-    from .kube.httpapi import serve_store
-    from .kube.store import APIStore
+    import json as jsonlib
 
-    serve_store(APIStore(), port)
+    from .kube.httpapi import APIServer
+    from .kube.store import APIStore
+    from .signals import setup_signal_handler
+
+    if state_file and os.path.exists(state_file):
+        with open(state_file) as f:
+            store = APIStore.load(jsonlib.load(f))
+        logger.info("Restored state from %s", state_file)
+    else:
+        store = APIStore()
+
+    server = APIServer(store, port, host="")
+    server.start()
+    logger.info("API server listening on :%d", server.port)
+    stop = setup_signal_handler()
+    stop.wait()
+    if state_file:
+        tmp = state_file + ".tmp"
+        with open(tmp, "w") as f:
+            jsonlib.dump(store.dump(), f)
+        os.replace(tmp, state_file)
+        logger.info("State saved to %s", state_file)
+    server.shutdown()
+
+
+@cli.command("get")
+@click.argument("kind")
+@click.option("-n", "--namespace", default=None, help="Namespace filter (default: all).")
+@click.option("--master", default="", help="agac API server URL (native wire scheme).")
+def get_cmd(kind, namespace, master):
+    """List objects of KIND (service/ingress/endpointgroupbinding/lease/event)
+    from an agac API server — kubectl-get for the embedded backends."""
+    from .kube.client import class_for_kind
+    from .kube.rest import RestKubeClient
+
+    aliases = {
+        "service": "Service", "services": "Service", "svc": "Service",
+        "ingress": "Ingress", "ingresses": "Ingress", "ing": "Ingress",
+        "endpointgroupbinding": "EndpointGroupBinding",
+        "endpointgroupbindings": "EndpointGroupBinding", "egb": "EndpointGroupBinding",
+        "lease": "Lease", "leases": "Lease",
+        "event": "Event", "events": "Event",
+    }
+    resolved = aliases.get(kind.lower())
+    if resolved is None:
+        raise click.UsageError(f"unknown kind {kind!r}; one of {sorted(set(aliases))}")
+    class_for_kind(resolved)  # validate early
+    server = master or os.environ.get("AGAC_API_SERVER", "http://127.0.0.1:8001")
+    client = RestKubeClient(server)
+    items, _ = client.list(resolved, namespace)
+    click.echo(f"{'NAMESPACE':<16} {'NAME':<40} {'RV':<8} DETAIL")
+    for obj in items:
+        detail = ""
+        if resolved == "EndpointGroupBinding":
+            detail = f"endpoints={len(obj.status.endpoint_ids)} gen={obj.metadata.generation} observed={obj.status.observed_generation}"
+        elif resolved == "Lease":
+            detail = f"holder={obj.spec.holder_identity}"
+        elif resolved == "Event":
+            detail = f"{obj.reason} x{obj.count}"
+        click.echo(
+            f"{obj.metadata.namespace:<16} {obj.metadata.name:<40} "
+            f"{obj.metadata.resource_version:<8} {detail}"
+        )
 
 
 def main():

@@ -324,6 +324,35 @@ class APIStore:
             self._broadcast(kind, DELETED, removed, rv)
             return None
 
+    # -- snapshot / restore -------------------------------------------------
+    def dump(self) -> dict:
+        """Serializable snapshot of every object + the rv counter (the
+        apiserver's checkpoint; controllers themselves are stateless and
+        resume from LIST, see docs/ARCHITECTURE.md)."""
+        with self._lock:
+            objects = [
+                {"kind": kind, "object": metalib.to_dict(obj)}
+                for kind, bucket in self._objects.items()
+                for obj in bucket.values()
+            ]
+            # peek the counter without burning (itertools.count has no peek:
+            # burn one and record it as the floor for the restored counter)
+            next_rv = next(self._rv)
+            return {"version": 1, "nextResourceVersion": next_rv, "objects": objects}
+
+    @classmethod
+    def load(cls, snapshot: dict) -> "APIStore":
+        from .client import class_for_kind
+
+        store = cls()
+        store._rv = itertools.count(int(snapshot.get("nextResourceVersion", 1)))
+        for entry in snapshot.get("objects", []):
+            obj_cls = class_for_kind(entry["kind"])
+            obj = metalib.from_dict(obj_cls, entry["object"])
+            key = (obj.metadata.namespace, obj.metadata.name)
+            store._bucket(entry["kind"])[key] = obj
+        return store
+
     # -- watch -------------------------------------------------------------
     def watch(
         self,
