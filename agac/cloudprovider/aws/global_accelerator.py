@@ -217,9 +217,16 @@ class GlobalAcceleratorMixin:
 
     # -- ensure (reference :112-211) ---------------------------------------
     def ensure_global_accelerator_for_service(
-        self, svc, lb_ingress, cluster_name: str, lb_name: str, region: str
+        self, svc, lb_ingress, cluster_name: str, lb_name: str, region: str,
+        hint_arn: Optional[str] = None,
     ):
-        """Returns (accelerator_arn | None, created, retry_after_seconds)."""
+        """Returns (accelerator_arn | None, created, retry_after_seconds).
+
+        ``hint_arn`` short-circuits the O(#accelerators × ListTags) discovery
+        scan (reference SURVEY §3.2 hot loop): the hinted accelerator is
+        fetched directly and its ownership tags VERIFIED before use; any
+        mismatch or error falls back to the full scan, so behavior is
+        identical — only steady-state API cost drops to O(1)."""
         return self._ensure_global_accelerator(
             obj=svc,
             resource="service",
@@ -230,10 +237,12 @@ class GlobalAcceleratorMixin:
             listener_spec=listener_for_service,
             protocol_changed=listener_protocol_changed_from_service,
             port_changed=listener_port_changed_from_service,
+            hint_arn=hint_arn,
         )
 
     def ensure_global_accelerator_for_ingress(
-        self, ingress, lb_ingress, cluster_name: str, lb_name: str, region: str
+        self, ingress, lb_ingress, cluster_name: str, lb_name: str, region: str,
+        hint_arn: Optional[str] = None,
     ):
         """Same as the service path, with ingress listener derivation.
         Unlike the reference (:241-244) a listener-create failure here is an
@@ -248,7 +257,31 @@ class GlobalAcceleratorMixin:
             listener_spec=listener_for_ingress,
             protocol_changed=listener_protocol_changed_from_ingress,
             port_changed=listener_port_changed_from_ingress,
+            hint_arn=hint_arn,
         )
+
+    def _verified_hint(
+        self, hint_arn: str, cluster_name: str, resource: str, ns: str, name: str
+    ):
+        """Fetch the hinted accelerator and verify our ownership tags;
+        returns [accelerator] or None to force the full discovery scan."""
+        try:
+            accelerator = self._get_accelerator(hint_arn)
+            tags = self._list_tags_for_accelerator(hint_arn)
+        except Exception:
+            return None
+        if tags_contains_all_values(
+            tags,
+            {
+                GLOBAL_ACCELERATOR_MANAGED_TAG_KEY: "true",
+                GLOBAL_ACCELERATOR_OWNER_TAG_KEY: accelerator_owner_tag_value(
+                    resource, ns, name
+                ),
+                GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY: cluster_name,
+            },
+        ):
+            return [accelerator]
+        return None
 
     def _ensure_global_accelerator(
         self,
@@ -261,6 +294,7 @@ class GlobalAcceleratorMixin:
         listener_spec,
         protocol_changed,
         port_changed,
+        hint_arn: Optional[str] = None,
     ):
         lb = self.get_load_balancer(lb_name)
         if lb.dns_name != hostname:
@@ -272,9 +306,15 @@ class GlobalAcceleratorMixin:
             return None, False, self.lb_not_active_retry
 
         logger.info("LoadBalancer is %s", lb.load_balancer_arn)
-        accelerators = self.list_global_accelerator_by_resource(
-            cluster_name, resource, obj.metadata.namespace, obj.metadata.name
-        )
+        accelerators = None
+        if hint_arn:
+            accelerators = self._verified_hint(
+                hint_arn, cluster_name, resource, obj.metadata.namespace, obj.metadata.name
+            )
+        if accelerators is None:
+            accelerators = self.list_global_accelerator_by_resource(
+                cluster_name, resource, obj.metadata.namespace, obj.metadata.name
+            )
         if not accelerators:
             logger.info("Creating Global Accelerator for %s", lb.dns_name)
             created_arn = self._create_global_accelerator(

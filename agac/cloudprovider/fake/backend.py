@@ -130,6 +130,7 @@ class FakeGlobalAccelerator:
     def __init__(self, lock: threading.RLock, deploy_after_describes: int = 1):
         self._lock = lock
         self.deploy_after_describes = deploy_after_describes
+        self.call_counts: Dict[str, int] = {}
         self._accelerators: Dict[str, t.Accelerator] = {}
         self._pending: Dict[str, int] = {}  # arn -> remaining IN_PROGRESS describes
         self._tags: Dict[str, Dict[str, str]] = {}
@@ -137,6 +138,9 @@ class FakeGlobalAccelerator:
         self._listener_owner: Dict[str, str] = {}  # listener arn -> accelerator arn
         self._endpoint_groups: Dict[str, t.EndpointGroup] = {}
         self._eg_owner: Dict[str, str] = {}  # endpoint group arn -> listener arn
+
+    def _count(self, op: str):
+        self.call_counts[op] = self.call_counts.get(op, 0) + 1
 
     # -- lifecycle helper --------------------------------------------------
     def _mutated(self, arn: str):
@@ -152,6 +156,7 @@ class FakeGlobalAccelerator:
         tags: Optional[List[t.Tag]] = None,
     ) -> t.Accelerator:
         with self._lock:
+            self._count('create_accelerator')
             arn = f"arn:aws:globalaccelerator::{_ACCOUNT}:accelerator/{uuid.uuid4()}"
             acc = t.Accelerator(
                 accelerator_arn=arn,
@@ -168,6 +173,7 @@ class FakeGlobalAccelerator:
 
     def describe_accelerator(self, arn: str) -> t.Accelerator:
         with self._lock:
+            self._count('describe_accelerator')
             acc = self._accelerators.get(arn)
             if acc is None:
                 raise awserr.AcceleratorNotFoundException(arn)
@@ -180,6 +186,7 @@ class FakeGlobalAccelerator:
 
     def list_accelerators(self, max_results: Optional[int] = None, next_token=None):
         with self._lock:
+            self._count('list_accelerators')
             items = sorted(self._accelerators.values(), key=lambda a: a.accelerator_arn)
             page, token = _paginate(items, max_results, next_token)
             return [_copy(a) for a in page], token
@@ -192,6 +199,7 @@ class FakeGlobalAccelerator:
         ip_address_type: Optional[str] = None,
     ) -> t.Accelerator:
         with self._lock:
+            self._count('update_accelerator')
             acc = self._accelerators.get(arn)
             if acc is None:
                 raise awserr.AcceleratorNotFoundException(arn)
@@ -206,6 +214,7 @@ class FakeGlobalAccelerator:
 
     def delete_accelerator(self, arn: str):
         with self._lock:
+            self._count('delete_accelerator')
             acc = self._accelerators.get(arn)
             if acc is None:
                 raise awserr.AcceleratorNotFoundException(arn)
@@ -224,12 +233,14 @@ class FakeGlobalAccelerator:
 
     def list_tags_for_resource(self, arn: str) -> List[t.Tag]:
         with self._lock:
+            self._count('list_tags_for_resource')
             if arn not in self._tags:
                 raise awserr.AcceleratorNotFoundException(arn)
             return [t.Tag(k, v) for k, v in self._tags[arn].items()]
 
     def tag_resource(self, arn: str, tags: List[t.Tag]):
         with self._lock:
+            self._count('tag_resource')
             if arn not in self._tags:
                 raise awserr.AcceleratorNotFoundException(arn)
             self._tags[arn].update({tag.key: tag.value for tag in tags})
@@ -243,6 +254,7 @@ class FakeGlobalAccelerator:
         client_affinity: str = t.CLIENT_AFFINITY_NONE,
     ) -> t.Listener:
         with self._lock:
+            self._count('create_listener')
             if accelerator_arn not in self._accelerators:
                 raise awserr.AcceleratorNotFoundException(accelerator_arn)
             arn = f"{accelerator_arn}/listener/{_short()}"
@@ -259,6 +271,7 @@ class FakeGlobalAccelerator:
 
     def list_listeners(self, accelerator_arn: str, max_results=None, next_token=None):
         with self._lock:
+            self._count('list_listeners')
             if accelerator_arn not in self._accelerators:
                 raise awserr.AcceleratorNotFoundException(accelerator_arn)
             items = sorted(
@@ -279,6 +292,7 @@ class FakeGlobalAccelerator:
         client_affinity: Optional[str] = None,
     ) -> t.Listener:
         with self._lock:
+            self._count('update_listener')
             listener = self._listeners.get(listener_arn)
             if listener is None:
                 raise awserr.ListenerNotFoundException(listener_arn)
@@ -293,6 +307,7 @@ class FakeGlobalAccelerator:
 
     def delete_listener(self, listener_arn: str):
         with self._lock:
+            self._count('delete_listener')
             if listener_arn not in self._listeners:
                 raise awserr.ListenerNotFoundException(listener_arn)
             owned = [e for e, l in self._eg_owner.items() if l == listener_arn]
@@ -314,6 +329,7 @@ class FakeGlobalAccelerator:
         endpoint_configurations: Optional[List[t.EndpointConfiguration]] = None,
     ) -> t.EndpointGroup:
         with self._lock:
+            self._count('create_endpoint_group')
             if listener_arn not in self._listeners:
                 raise awserr.ListenerNotFoundException(listener_arn)
             arn = f"{listener_arn}/endpoint-group/{_short()}"
@@ -331,6 +347,7 @@ class FakeGlobalAccelerator:
 
     def list_endpoint_groups(self, listener_arn: str, max_results=None, next_token=None):
         with self._lock:
+            self._count('list_endpoint_groups')
             if listener_arn not in self._listeners:
                 raise awserr.ListenerNotFoundException(listener_arn)
             items = sorted(e for e, l in self._eg_owner.items() if l == listener_arn)
@@ -339,6 +356,7 @@ class FakeGlobalAccelerator:
 
     def describe_endpoint_group(self, endpoint_group_arn: str) -> t.EndpointGroup:
         with self._lock:
+            self._count('describe_endpoint_group')
             eg = self._endpoint_groups.get(endpoint_group_arn)
             if eg is None:
                 raise awserr.EndpointGroupNotFoundException(endpoint_group_arn)
@@ -351,6 +369,7 @@ class FakeGlobalAccelerator:
     ) -> t.EndpointGroup:
         """Real-AWS semantics: EndpointConfigurations REPLACES the whole set."""
         with self._lock:
+            self._count('update_endpoint_group')
             eg = self._endpoint_groups.get(endpoint_group_arn)
             if eg is None:
                 raise awserr.EndpointGroupNotFoundException(endpoint_group_arn)
@@ -367,6 +386,7 @@ class FakeGlobalAccelerator:
         endpoint_configurations: List[t.EndpointConfiguration],
     ) -> List[t.EndpointDescription]:
         with self._lock:
+            self._count('add_endpoints')
             eg = self._endpoint_groups.get(endpoint_group_arn)
             if eg is None:
                 raise awserr.EndpointGroupNotFoundException(endpoint_group_arn)
@@ -394,6 +414,7 @@ class FakeGlobalAccelerator:
 
     def remove_endpoints(self, endpoint_group_arn: str, endpoint_ids: List[str]):
         with self._lock:
+            self._count('remove_endpoints')
             eg = self._endpoint_groups.get(endpoint_group_arn)
             if eg is None:
                 raise awserr.EndpointGroupNotFoundException(endpoint_group_arn)
@@ -404,6 +425,7 @@ class FakeGlobalAccelerator:
 
     def delete_endpoint_group(self, endpoint_group_arn: str):
         with self._lock:
+            self._count('delete_endpoint_group')
             if endpoint_group_arn not in self._endpoint_groups:
                 raise awserr.EndpointGroupNotFoundException(endpoint_group_arn)
             listener_arn = self._eg_owner.pop(endpoint_group_arn)

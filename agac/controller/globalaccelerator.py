@@ -54,6 +54,12 @@ class GlobalAcceleratorController:
         self.cluster_name = config.cluster_name
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
+        # (resource, ns, name) -> accelerator ARN hint.  Purely an API-cost
+        # optimization: the resource manager re-verifies ownership tags
+        # before trusting a hint, and any miss falls back to the full
+        # ListAccelerators scan (see ensure_global_accelerator_for_service).
+        self._arn_hints = {}
+        self._hints_lock = threading.Lock()
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
         self.service_queue = RateLimitingQueue(
             rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
@@ -173,12 +179,23 @@ class GlobalAcceleratorController:
         return reconcile.Result()
 
     def _cleanup_accelerators(self, resource: str, ns: str, name: str):
+        with self._hints_lock:
+            self._arn_hints.pop((resource, ns, name), None)
         cloud = self.cloud_factory("us-west-2")
         accelerators = cloud.list_global_accelerator_by_resource(
             self.cluster_name, resource, ns, name
         )
         for accelerator in accelerators:
             cloud.cleanup_global_accelerator(accelerator.accelerator_arn)
+
+    def _hint_for(self, resource: str, ns: str, name: str):
+        with self._hints_lock:
+            return self._arn_hints.get((resource, ns, name))
+
+    def _remember_hint(self, resource: str, ns: str, name: str, arn):
+        with self._hints_lock:
+            if arn:
+                self._arn_hints[(resource, ns, name)] = arn
 
     def process_service_create_or_update(self, svc) -> reconcile.Result:
         if len(svc.status.load_balancer.ingress) < 1:
@@ -215,7 +232,13 @@ class GlobalAcceleratorController:
             name, region = get_lb_name_from_hostname(lb_ingress.hostname)
             cloud = self.cloud_factory(region)
             arn, created, retry_after = cloud.ensure_global_accelerator_for_service(
-                svc, lb_ingress, self.cluster_name, name, region
+                svc, lb_ingress, self.cluster_name, name, region,
+                hint_arn=self._hint_for(
+                    "service", svc.metadata.namespace, svc.metadata.name
+                ),
+            )
+            self._remember_hint(
+                "service", svc.metadata.namespace, svc.metadata.name, arn
             )
             if retry_after > 0:
                 return reconcile.Result(requeue=True, requeue_after=retry_after)
@@ -274,7 +297,13 @@ class GlobalAcceleratorController:
             name, region = get_lb_name_from_hostname(lb_ingress.hostname)
             cloud = self.cloud_factory(region)
             arn, created, retry_after = cloud.ensure_global_accelerator_for_ingress(
-                ingress, lb_ingress, self.cluster_name, name, region
+                ingress, lb_ingress, self.cluster_name, name, region,
+                hint_arn=self._hint_for(
+                    "ingress", ingress.metadata.namespace, ingress.metadata.name
+                ),
+            )
+            self._remember_hint(
+                "ingress", ingress.metadata.namespace, ingress.metadata.name, arn
             )
             if retry_after > 0:
                 return reconcile.Result(requeue=True, requeue_after=retry_after)
