@@ -74,8 +74,12 @@ class Route53Mixin:
     """Methods bound into ``agac.cloudprovider.aws.client.AWS``."""
 
     def ensure_route53_for_service(
-        self, svc, lb_ingress, hostnames: List[str], cluster_name: str
+        self, svc, lb_ingress, hostnames: List[str], cluster_name: str,
+        hint_arn: Optional[str] = None,
     ) -> Tuple[bool, float]:
+        """``hint_arn`` short-circuits the by-hostname accelerator scan with
+        tag-verified fallback (same contract as
+        ensure_global_accelerator_for_service's hint)."""
         return self._ensure_route53(
             lb_ingress.hostname,
             hostnames,
@@ -83,10 +87,12 @@ class Route53Mixin:
             "service",
             svc.metadata.namespace,
             svc.metadata.name,
+            hint_arn=hint_arn,
         )
 
     def ensure_route53_for_ingress(
-        self, ingress, lb_ingress, hostnames: List[str], cluster_name: str
+        self, ingress, lb_ingress, hostnames: List[str], cluster_name: str,
+        hint_arn: Optional[str] = None,
     ) -> Tuple[bool, float]:
         return self._ensure_route53(
             lb_ingress.hostname,
@@ -95,7 +101,32 @@ class Route53Mixin:
             "ingress",
             ingress.metadata.namespace,
             ingress.metadata.name,
+            hint_arn=hint_arn,
         )
+
+    def _verified_hostname_hint(self, hint_arn: str, lb_hostname: str, cluster_name: str):
+        from .global_accelerator import (
+            GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY,
+            GLOBAL_ACCELERATOR_MANAGED_TAG_KEY,
+            GLOBAL_ACCELERATOR_TARGET_HOSTNAME_KEY,
+            tags_contains_all_values,
+        )
+
+        try:
+            accelerator = self._get_accelerator(hint_arn)
+            tags = self._list_tags_for_accelerator(hint_arn)
+        except Exception:
+            return None
+        if tags_contains_all_values(
+            tags,
+            {
+                GLOBAL_ACCELERATOR_MANAGED_TAG_KEY: "true",
+                GLOBAL_ACCELERATOR_TARGET_HOSTNAME_KEY: lb_hostname,
+                GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY: cluster_name,
+            },
+        ):
+            return [accelerator]
+        return None
 
     def _ensure_route53(
         self,
@@ -105,12 +136,19 @@ class Route53Mixin:
         resource: str,
         ns: str,
         name: str,
+        hint_arn: Optional[str] = None,
     ) -> Tuple[bool, float]:
         """Returns (created, retry_after_seconds).  0 or >1 matching
         accelerators ⇒ requeue after 60s (reference :62-78)."""
-        accelerators = self.list_global_accelerator_by_hostname(
-            lb_hostname, cluster_name
-        )
+        accelerators = None
+        if hint_arn:
+            accelerators = self._verified_hostname_hint(
+                hint_arn, lb_hostname, cluster_name
+            )
+        if accelerators is None:
+            accelerators = self.list_global_accelerator_by_hostname(
+                lb_hostname, cluster_name
+            )
         if len(accelerators) > 1:
             logger.error("Too many Global Accelerators for %s", lb_hostname)
             return False, self.ga_missing_retry
@@ -118,6 +156,9 @@ class Route53Mixin:
             logger.error("Could not find Global Accelerator for %s", lb_hostname)
             return False, self.ga_missing_retry
         accelerator = accelerators[0]
+        # expose the matched accelerator so callers can maintain their hint
+        # without a second scan (read via getattr right after the call)
+        self._last_matched_accelerator_arn = accelerator.accelerator_arn
 
         owner_value = route53_owner_value(cluster_name, resource, ns, name)
         created = False

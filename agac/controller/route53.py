@@ -54,6 +54,10 @@ class Route53Controller:
         self.cluster_name = config.cluster_name
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
+        # LB hostname -> accelerator ARN hint (tag-verified before use;
+        # misses fall back to the full by-hostname scan)
+        self._arn_hints = {}
+        self._hints_lock = threading.Lock()
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
         self.service_queue = RateLimitingQueue(
             rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
@@ -116,6 +120,19 @@ class Route53Controller:
 
     def _enqueue_ingress(self, obj):
         self.ingress_queue.add_rate_limited(meta_namespace_key(obj))
+
+    def _hint_for(self, lb_hostname: str):
+        with self._hints_lock:
+            return self._arn_hints.get(lb_hostname)
+
+    def _remember_hint(self, lb_hostname: str, cloud):
+        """Record the accelerator the ensure call actually matched (exposed
+        by the resource manager) — seeds and self-heals the hint without
+        any extra API call."""
+        arn = getattr(cloud, "_last_matched_accelerator_arn", None)
+        if arn:
+            with self._hints_lock:
+                self._arn_hints[lb_hostname] = arn
 
     # -- run ----------------------------------------------------------------
     def run(self, threadiness: int, stop: threading.Event):
@@ -196,8 +213,11 @@ class Route53Controller:
             _, region = get_lb_name_from_hostname(lb_ingress.hostname)
             cloud = self.cloud_factory(region)
             created, retry_after = cloud.ensure_route53_for_service(
-                svc, lb_ingress, hostnames, self.cluster_name
+                svc, lb_ingress, hostnames, self.cluster_name,
+                hint_arn=self._hint_for(lb_ingress.hostname),
             )
+            if retry_after == 0:
+                self._remember_hint(lb_ingress.hostname, cloud)
             if retry_after > 0:
                 return reconcile.Result(requeue=True, requeue_after=retry_after)
             if created:
@@ -257,8 +277,11 @@ class Route53Controller:
             _, region = get_lb_name_from_hostname(lb_ingress.hostname)
             cloud = self.cloud_factory(region)
             created, retry_after = cloud.ensure_route53_for_ingress(
-                ingress, lb_ingress, hostnames, self.cluster_name
+                ingress, lb_ingress, hostnames, self.cluster_name,
+                hint_arn=self._hint_for(lb_ingress.hostname),
             )
+            if retry_after == 0:
+                self._remember_hint(lb_ingress.hostname, cloud)
             if retry_after > 0:
                 return reconcile.Result(requeue=True, requeue_after=retry_after)
             if created:
