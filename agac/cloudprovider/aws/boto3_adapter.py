@@ -340,7 +340,8 @@ class Boto3Route53:
         res = _call(self.client.list_hosted_zones_by_name, **kwargs)
         return [self._zone_out(z) for z in res.get("HostedZones", [])]
 
-    def list_resource_record_sets(self, zone_id, max_items=None, start_token=None):
+    def list_resource_record_sets(self, zone_id, max_items=None, start_token=None,
+                                  start_record_name=None):
         kwargs = {"HostedZoneId": zone_id}
         if max_items:
             kwargs["MaxItems"] = str(max_items)
@@ -348,6 +349,8 @@ class Boto3Route53:
             name, rtype = start_token.split("|", 1)
             kwargs["StartRecordName"] = name
             kwargs["StartRecordType"] = rtype
+        elif start_record_name:
+            kwargs["StartRecordName"] = start_record_name
         res = _call(self.client.list_resource_record_sets, **kwargs)
         records = [self._record_out(r) for r in res.get("ResourceRecordSets", [])]
         token = None

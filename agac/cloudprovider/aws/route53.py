@@ -165,8 +165,7 @@ class Route53Mixin:
         for hostname in hostnames:
             hosted_zone = self.get_hosted_zone(hostname)
             logger.info("HostedZone is %s", hosted_zone.id)
-            records = self.find_owned_a_record_sets(hosted_zone, owner_value)
-            record = find_a_record(records, hostname)
+            record = self._find_owned_a_record_at(hosted_zone, hostname, owner_value)
             if record is None:
                 logger.info(
                     "Creating record for %s with %s",
@@ -203,6 +202,51 @@ class Route53Mixin:
                 logger.info("Record set %s: %s is deleted", record.name, record.type)
 
     # -- record discovery ---------------------------------------------------
+    def _find_owned_a_record_at(self, hosted_zone, hostname: str, owner_value: str):
+        """Name-scoped variant of the reference's owned-record discovery
+        (FindOwneredARecordSets + findARecord scan the whole zone,
+        route53.go:216-238/360-367): list only the records AT ``hostname``
+        via StartRecordName and decide from those.  The decision — create
+        (no owned TXT), skip/update (owned TXT + alias A) — is identical;
+        only the API cost drops from O(zone records) to O(1).  The cleanup
+        paths still use the full-zone scan (they must find every owned
+        name)."""
+        target = hostname + "." if not hostname.endswith(".") else hostname
+        target = target.replace("*", "\\052", 1)
+        records_at_name: List[t.ResourceRecordSet] = []
+        token = None
+        while True:
+            metrics.observe_aws_call("route53", "ListResourceRecordSets")
+            page, token = self.route53.list_resource_record_sets(
+                hosted_zone.id, max_items=10, start_token=token,
+                start_record_name=hostname,
+            )
+            done = False
+            for rs in page:
+                if rs.name == target:
+                    records_at_name.append(rs)
+                elif rs.name > target:
+                    done = True
+                    break
+            if done or token is None:
+                break
+        owned = any(
+            record.value == owner_value
+            for rs in records_at_name
+            for record in rs.resource_records
+        )
+        if not owned:
+            return None
+        return next(
+            (
+                rs
+                for rs in records_at_name
+                if rs.alias_target is not None
+                and replace_wildcards(rs.name) == hostname + "."
+            ),
+            None,
+        )
+
     def find_owned_a_record_sets(
         self, hosted_zone: t.HostedZone, owner_value: str
     ) -> List[t.ResourceRecordSet]:

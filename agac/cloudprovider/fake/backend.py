@@ -480,12 +480,19 @@ class FakeRoute53:
             page, _ = _paginate(after, max_items, None)
             return [_copy(z) for z in page]
 
-    def list_resource_record_sets(self, zone_id: str, max_items=None, start_token=None):
+    def list_resource_record_sets(
+        self, zone_id: str, max_items=None, start_token=None, start_record_name=None
+    ):
+        """start_record_name mirrors the real API's StartRecordName: begin
+        at the first record whose name >= the (normalized) given name."""
         with self._lock:
             records = self._records.get(zone_id)
             if records is None:
                 raise awserr.NoSuchHostedZone(zone_id)
             items = [records[k] for k in sorted(records)]
+            if start_record_name is not None:
+                start = _normalize_record_name(start_record_name)
+                items = [r for r in items if r.name >= start]
             page, token = _paginate(items, max_items, start_token)
             return [_copy(r) for r in page], token
 
