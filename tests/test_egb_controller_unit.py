@@ -94,7 +94,7 @@ class TestNotifications:
         new.metadata.annotations["x"] = "y"
         controller._update_notification(old, new)
         # rate-limited add lands after a few ms
-        item, shutdown = controller.workqueue.get(timeout=2.0)
+        item, shutdown = controller.workqueue.get(timeout=10.0)
         assert item == "default/b" and not shutdown
 
 

@@ -53,7 +53,7 @@ class TestWatchLogExpiry:
         _, rv = store.list("Service")
         watch = store.watch("Service", resource_version=rv)
         store.create(mk_service("second"))
-        event = watch.get(timeout=1.0)
+        event = watch.get(timeout=10.0)
         assert event.obj.metadata.name == "second"
         watch.stop()
 
@@ -125,6 +125,6 @@ class TestStoreFieldImmutability:
         watch = store.watch("Service", namespace="ns1", resource_version=rv)
         store.create(mk_service("other", ns="ns2"))
         store.create(mk_service("mine", ns="ns1"))
-        event = watch.get(timeout=1.0)
+        event = watch.get(timeout=10.0)
         assert event.obj.metadata.name == "mine"
         watch.stop()

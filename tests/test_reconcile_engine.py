@@ -68,7 +68,7 @@ def test_error_requeues_rate_limited():
 
     pump(q, lambda k: Obj("x"), lambda k: Result(), boom)
     q.done("default/x")
-    item, shutdown = q.get(timeout=2.0)
+    item, shutdown = q.get(timeout=10.0)
     assert item == "default/x" and not shutdown
     assert q.num_requeues("default/x") >= 1
 
@@ -97,7 +97,7 @@ def test_requeue_after_uses_add_after_and_forgets():
     pump(q, lambda k: Obj("x"), lambda k: Result(), process)
     q.done("default/x")
     assert q.num_requeues("default/x") == 0  # forgotten before AddAfter
-    item, _ = q.get(timeout=2.0)
+    item, _ = q.get(timeout=10.0)
     assert item == "default/x"
 
 
@@ -105,7 +105,7 @@ def test_requeue_flag_rate_limits():
     q = make_queue()
     pump(q, lambda k: Obj("x"), lambda k: Result(), lambda o: Result(requeue=True))
     q.done("default/x")
-    item, _ = q.get(timeout=2.0)
+    item, _ = q.get(timeout=10.0)
     assert item == "default/x"
     assert q.num_requeues("default/x") >= 1
 

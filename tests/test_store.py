@@ -125,7 +125,7 @@ def test_list_and_watch_replay():
     assert [o.metadata.name for o in items] == ["a"]
     w = s.watch("Service", resource_version=rv)
     s.create(mk_service("b"))
-    ev = w.get(timeout=1.0)
+    ev = w.get(timeout=10.0)
     assert ev.type == "ADDED" and ev.obj.metadata.name == "b"
     w.stop()
 
@@ -138,10 +138,10 @@ def test_watch_sees_modify_and_delete():
     obj = s.get("Service", "default", "web")
     obj.metadata.annotations["k"] = "v"
     s.update(obj)
-    ev = w.get(timeout=1.0)
+    ev = w.get(timeout=10.0)
     assert ev.type == "MODIFIED"
     s.delete("Service", "default", "web")
-    ev = w.get(timeout=1.0)
+    ev = w.get(timeout=10.0)
     assert ev.type == "DELETED"
     w.stop()
 

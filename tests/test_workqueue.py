@@ -55,7 +55,7 @@ def test_add_after_delivers():
     q = make_queue()
     q.add_after("x", 0.02)
     assert len(q) == 0
-    item, shutdown = q.get(timeout=2.0)
+    item, shutdown = q.get(timeout=10.0)
     assert item == "x" and not shutdown
 
 
@@ -63,9 +63,9 @@ def test_add_after_ordering():
     q = make_queue()
     q.add_after("late", 0.05)
     q.add_after("early", 0.005)
-    first, _ = q.get(timeout=2.0)
+    first, _ = q.get(timeout=10.0)
     assert first == "early"
-    second, _ = q.get(timeout=2.0)
+    second, _ = q.get(timeout=10.0)
     assert second == "late"
 
 
@@ -126,7 +126,7 @@ def test_shutdown_wakes_blocked_worker():
     t.start()
     time.sleep(0.05)
     q.shut_down()
-    t.join(timeout=2.0)
+    t.join(timeout=10.0)
     assert not t.is_alive()
     assert results == [(None, True)]
 
