@@ -265,6 +265,102 @@ def get_cmd(kind, namespace, master):
         )
 
 
+@cli.command()
+@click.option("--objects", default=3, show_default=True, help="Sample services to reconcile.")
+def demo(objects):
+    """Guided demo: run the full stack in-process (embedded API + AWS fake),
+    reconcile sample services into Global Accelerators + Route53 records,
+    then tear one down — printing the resulting cloud state and events."""
+    import threading
+    import time
+
+    from .apis import core as corev1
+    from .apis.meta import ObjectMeta
+    from .cloudprovider.aws.client import FakeCloudFactory
+    from .cloudprovider.fake import FakeAWSBackend
+    from .kube.client import InMemoryKubeClient
+    from .manager import ControllerConfig, Manager
+
+    MANAGED = "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-managed"
+    HOSTNAME = "aws-global-accelerator-controller.h3poteto.dev/route53-hostname"
+    LB_TYPE = "service.beta.kubernetes.io/aws-load-balancer-type"
+
+    client = InMemoryKubeClient()
+    backend = FakeAWSBackend()
+    backend.route53.create_hosted_zone("demo.example.com")
+    stop = threading.Event()
+    manager = Manager()
+    manager.run(
+        client, ControllerConfig(),
+        FakeCloudFactory(backend, ga_missing_retry=0.1),
+        stop, resync_period=5.0, block=False,
+    )
+    manager.wait_until_ready()
+    click.echo(f"controllers up; creating {objects} LoadBalancer Services "
+               "with managed + route53 annotations...")
+    try:
+        for i in range(objects):
+            lb = backend.elbv2.create_load_balancer(f"demo-{i}", region="us-east-1")
+            client.create(
+                corev1.Service(
+                    metadata=ObjectMeta(
+                        name=f"demo-{i}", namespace="default",
+                        annotations={
+                            LB_TYPE: "nlb", MANAGED: "true",
+                            HOSTNAME: f"demo-{i}.demo.example.com",
+                        },
+                    ),
+                    spec=corev1.ServiceSpec(
+                        type="LoadBalancer",
+                        ports=[corev1.ServicePort(port=80, protocol="TCP")],
+                    ),
+                    status=corev1.ServiceStatus(
+                        load_balancer=corev1.LoadBalancerStatus(
+                            ingress=[corev1.LoadBalancerIngress(hostname=lb.dns_name)]
+                        )
+                    ),
+                )
+            )
+
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            accs, _ = backend.ga.list_accelerators()
+            zones, _ = backend.route53.list_hosted_zones()
+            records, _ = backend.route53.list_resource_record_sets(zones[0].id)
+            if len(accs) == objects and len(records) == 2 * objects:
+                break
+            time.sleep(0.05)
+
+        click.echo("\nGlobal Accelerators:")
+        for acc in accs:
+            listeners, _ = backend.ga.list_listeners(acc.accelerator_arn)
+            ports = [p.from_port for p in listeners[0].port_ranges]
+            click.echo(f"  {acc.name:<24} {acc.dns_name:<44} ports={ports}")
+        click.echo("\nRoute53 records (demo.example.com):")
+        for record in records:
+            target = record.alias_target.dns_name if record.alias_target else \
+                record.resource_records[0].value[:40] + "..."
+            click.echo(f"  {record.type:<4} {record.name:<34} -> {target}")
+
+        click.echo(f"\ndeleting Service demo-0 ...")
+        client.delete("Service", "default", "demo-0")
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            accs, _ = backend.ga.list_accelerators()
+            if len(accs) == objects - 1:
+                break
+            time.sleep(0.05)
+        click.echo(f"accelerators after delete: {len(accs)} (was {objects})")
+
+        click.echo("\nEvents:")
+        events, _ = client.list("Event")
+        for ev in events:
+            click.echo(f"  {ev.involved_object.name:<10} {ev.reason:<26} x{ev.count}")
+    finally:
+        stop.set()
+    click.echo("\ndemo OK")
+
+
 def main():
     cli()
 

@@ -80,3 +80,13 @@ class TestFixture:
         binding = endpoint_group_binding(ingress_name="my-ingress")
         assert binding.spec.ingress_ref.name == "my-ingress"
         assert binding.spec.service_ref is None
+
+
+class TestDemo:
+    def test_demo_runs_end_to_end(self):
+        from click.testing import CliRunner
+
+        result = CliRunner().invoke(cli, ["demo", "--objects", "2"])
+        assert result.exit_code == 0, result.output
+        assert "demo OK" in result.output
+        assert "GlobalAcceleratorCreated" in result.output
