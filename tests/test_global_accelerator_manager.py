@@ -349,3 +349,38 @@ class TestDeleteAcceleratorLifecycle:
 
         with pytest.raises(AcceleratorNotDisabledException):
             backend.ga.delete_accelerator(acc.accelerator_arn)
+
+
+class TestUserTagDrift:
+    def test_tags_annotation_change_retags_accelerator(self, backend, cloud):
+        svc = mk_service()
+        lb = seed_lb(backend)
+        arn, _, _ = cloud.ensure_global_accelerator_for_service(
+            svc, lb_ingress(lb), "c", "mylb", REGION
+        )
+        # user adds the tags annotation later → accelerator must be retagged
+        svc.metadata.annotations[
+            "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-tags"
+        ] = "env=prod,team=net"
+        cloud.ensure_global_accelerator_for_service(
+            svc, lb_ingress(lb), "c", "mylb", REGION
+        )
+        tags = {x.key: x.value for x in backend.ga.list_tags_for_resource(arn)}
+        assert tags["env"] == "prod"
+        assert tags["team"] == "net"
+        # ownership tags survive the retag
+        assert tags["aws-global-accelerator-owner"] == "service/default/web"
+
+    def test_name_annotation_change_renames(self, backend, cloud):
+        svc = mk_service()
+        lb = seed_lb(backend)
+        arn, _, _ = cloud.ensure_global_accelerator_for_service(
+            svc, lb_ingress(lb), "c", "mylb", REGION
+        )
+        svc.metadata.annotations[
+            "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-name"
+        ] = "renamed"
+        cloud.ensure_global_accelerator_for_service(
+            svc, lb_ingress(lb), "c", "mylb", REGION
+        )
+        assert backend.ga.describe_accelerator(arn).name == "renamed"
