@@ -30,24 +30,47 @@ from . import __version__
 logger = logging.getLogger(__name__)
 
 
-def _setup_logging(verbosity: int):
+class _JsonFormatter(logging.Formatter):
+    """One JSON object per line (for log pipelines)."""
+
+    def format(self, record):
+        import json as jsonlib
+
+        entry = {
+            "ts": self.formatTime(record, "%Y-%m-%dT%H:%M:%S"),
+            "level": record.levelname,
+            "logger": record.name,
+            "message": record.getMessage(),
+        }
+        if record.exc_info:
+            entry["exc"] = self.formatException(record.exc_info)
+        return jsonlib.dumps(entry)
+
+
+def _setup_logging(verbosity: int, log_format: str = "text"):
     level = logging.WARNING
     if verbosity == 1:
         level = logging.INFO
     elif verbosity >= 2:
         level = logging.DEBUG
-    logging.basicConfig(
-        level=level,
-        format="%(asctime)s %(levelname).1s %(name)s %(message)s",
-        stream=sys.stderr,
-    )
+    handler = logging.StreamHandler(sys.stderr)
+    if log_format == "json":
+        handler.setFormatter(_JsonFormatter())
+    else:
+        handler.setFormatter(
+            logging.Formatter("%(asctime)s %(levelname).1s %(name)s %(message)s")
+        )
+    root = logging.getLogger()
+    root.setLevel(level)
+    root.addHandler(handler)
 
 
 @click.group()
 @click.option("-v", "--verbose", count=True, help="Increase log verbosity (klog-style).")
-def cli(verbose: int):
+@click.option("--log-format", type=click.Choice(["text", "json"]), default="text", show_default=True)
+def cli(verbose: int, log_format: str):
     """aws-global-accelerator-controller (agac)."""
-    _setup_logging(verbose)
+    _setup_logging(verbose, log_format)
 
 
 def resolve_kubeconfig(kubeconfig: str) -> str:
