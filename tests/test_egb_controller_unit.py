@@ -208,3 +208,59 @@ class TestReconcileBranches:
         sync_informers(factory)
         res = controller.reconcile(client.get("EndpointGroupBinding", "default", "b"))
         assert res.requeue and res.requeue_after == 30.0
+
+
+class TestMultiRegionEndpoints:
+    def test_binding_attaches_lbs_across_regions(self, setup):
+        """BASELINE.json config 4: EndpointGroupBinding reconcile with
+        multi-region endpoints — a Service backed by LBs in two regions gets
+        both attached; delete drains each via its own ARN region."""
+        client, backend, controller, factory = setup
+        group = seed_group(backend)
+
+        lb_east = backend.elbv2.create_load_balancer("multi", region="us-east-1")
+        lb_west = backend.elbv2.create_load_balancer("multi", region="eu-west-1")
+        svc = corev1.Service(
+            metadata=ObjectMeta(name="multi", namespace="default"),
+            spec=corev1.ServiceSpec(type="LoadBalancer"),
+            status=corev1.ServiceStatus(
+                load_balancer=corev1.LoadBalancerStatus(
+                    ingress=[
+                        corev1.LoadBalancerIngress(hostname=lb_east.dns_name),
+                        corev1.LoadBalancerIngress(hostname=lb_west.dns_name),
+                    ]
+                )
+            ),
+        )
+        client.create(svc)
+        client.create(
+            mk_binding(group.endpoint_group_arn, name="mr", finalizers=[FINALIZER],
+                       service="multi")
+        )
+        sync_informers(factory)
+        controller.reconcile(client.get("EndpointGroupBinding", "default", "mr"))
+
+        stored = client.get("EndpointGroupBinding", "default", "mr")
+        assert sorted(stored.status.endpoint_ids) == sorted(
+            [lb_east.load_balancer_arn, lb_west.load_balancer_arn]
+        )
+        attached = {
+            d.endpoint_id
+            for d in backend.ga.describe_endpoint_group(
+                group.endpoint_group_arn
+            ).endpoint_descriptions
+        }
+        assert attached == {lb_east.load_balancer_arn, lb_west.load_balancer_arn}
+        # the two ARNs carry different regions (exercises per-ARN region
+        # client selection on the drain path)
+        from agac.cloudprovider.aws import get_region_from_arn
+
+        regions = {get_region_from_arn(a) for a in stored.status.endpoint_ids}
+        assert regions == {"us-east-1", "eu-west-1"}
+
+        # drain: one pass removes both
+        client.delete("EndpointGroupBinding", "default", "mr")
+        controller.reconcile(client.get("EndpointGroupBinding", "default", "mr"))
+        assert backend.ga.describe_endpoint_group(
+            group.endpoint_group_arn
+        ).endpoint_descriptions == []
