@@ -86,8 +86,10 @@ class FakeCloudFactory:
         self.ga_missing_retry = ga_missing_retry
 
     def __call__(self, region: str) -> AWS:
+        from ..fake.backend import RegionalELBv2View
+
         return AWS(
-            lb=self.backend.elbv2,
+            lb=RegionalELBv2View(self.backend.elbv2, region),
             ga=self.backend.ga,
             route53=self.backend.route53,
             region=region,

@@ -126,6 +126,37 @@ class FakeELBv2:
             return [_copy(lb) for lb in page], next_marker
 
 
+class RegionalELBv2View:
+    """Region-scoped view over the account-wide FakeELBv2 — real ELBv2
+    clients are regional, so a us-east-1 client must not see eu-west-1
+    load balancers (caught by the multi-region EndpointGroupBinding test)."""
+
+    def __init__(self, elbv2: FakeELBv2, region: str):
+        self._elbv2 = elbv2
+        self.region = region
+
+    def describe_load_balancers(self, names=None, marker=None, page_size=None):
+        with self._elbv2._lock:
+            lbs = sorted(
+                (
+                    lb
+                    for arn, lb in self._elbv2._lbs.items()
+                    if self._elbv2._regions.get(arn) == self.region
+                ),
+                key=lambda x: x.load_balancer_arn,
+            )
+            if names:
+                found = [lb for lb in lbs if lb.load_balancer_name in names]
+                missing = set(names) - {lb.load_balancer_name for lb in found}
+                if missing:
+                    raise awserr.LoadBalancerNotFoundException(
+                        f"Load balancers '[{', '.join(sorted(missing))}]' not found"
+                    )
+                return [_copy(lb) for lb in found], None
+            page, next_marker = _paginate(lbs, page_size, marker)
+            return [_copy(lb) for lb in page], next_marker
+
+
 class FakeGlobalAccelerator:
     def __init__(self, lock: threading.RLock, deploy_after_describes: int = 1):
         self._lock = lock
