@@ -6,6 +6,9 @@ PYTHON ?= python3
 test:
 	$(PYTHON) -m pytest tests/ -q -m "not gpu"
 
+test-fast:
+	$(PYTHON) -m pytest tests/ -q -m "not gpu" -n 4
+
 # threading stress profile: run the suite repeatedly with randomized order
 test-race:
 	$(PYTHON) -m pytest tests/ -q -m "not gpu" -p no:cacheprovider --count 3 2>/dev/null \

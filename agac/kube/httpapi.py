@@ -18,6 +18,12 @@ Wire scheme (generic, k8s-shaped):
       → ndjson stream of {"type", "object", "resourceVersion"}
 - ``GET    /healthz``
 
+The REAL Kubernetes REST surface is served too (``_k8s_route``):
+``/api/v1/namespaces/{ns}/services``, ``/apis/<group>/<version>/…``,
+``?watch=true`` streaming with apiserver event framing, ``<Kind>List``
+bodies and ``Status`` failure objects — so ``K8sKubeClient`` (and any
+client-go-shaped client) runs against this server unchanged.
+
 Errors: JSON ``{"code", "reason", "message"}`` with the matching HTTP
 status; reasons NotFound / AlreadyExists / Conflict / Gone map back to the
 typed store errors in the client.
