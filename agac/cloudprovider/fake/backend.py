@@ -51,6 +51,7 @@ class FakeELBv2:
         self._lock = lock
         self._lbs: Dict[str, t.LoadBalancer] = {}  # arn -> LB
         self._regions: Dict[str, str] = {}  # arn -> region
+        self._by_name: Dict[Tuple[str, str], str] = {}  # (region, name) -> arn
 
     # -- test/bench seeding helpers ---------------------------------------
     def create_load_balancer(
@@ -63,15 +64,11 @@ class FakeELBv2:
         dns_name: Optional[str] = None,
     ) -> t.LoadBalancer:
         with self._lock:
-            for existing_arn, existing in self._lbs.items():
-                if (
-                    existing.load_balancer_name == name
-                    and self._regions.get(existing_arn) == region
-                ):
-                    raise awserr.AWSAPIError(
-                        f"A load balancer with the name '{name}' already exists",
-                        "DuplicateLoadBalancerName",
-                    )
+            if (region, name) in self._by_name:
+                raise awserr.AWSAPIError(
+                    f"A load balancer with the name '{name}' already exists",
+                    "DuplicateLoadBalancerName",
+                )
             kind = "net" if lb_type == "network" else "app"
             arn = (
                 f"arn:aws:elasticloadbalancing:{region}:{_ACCOUNT}:"
@@ -94,6 +91,7 @@ class FakeELBv2:
             )
             self._lbs[arn] = lb
             self._regions[arn] = region
+            self._by_name[(region, name)] = arn
             return _copy(lb)
 
     def set_state(self, name_or_arn: str, state: str):
@@ -137,6 +135,19 @@ class RegionalELBv2View:
 
     def describe_load_balancers(self, names=None, marker=None, page_size=None):
         with self._elbv2._lock:
+            if names:
+                found, missing = [], []
+                for name in names:
+                    arn = self._elbv2._by_name.get((self.region, name))
+                    if arn is None:
+                        missing.append(name)
+                    else:
+                        found.append(self._elbv2._lbs[arn])
+                if missing:
+                    raise awserr.LoadBalancerNotFoundException(
+                        f"Load balancers '[{', '.join(sorted(missing))}]' not found"
+                    )
+                return [_copy(lb) for lb in found], None
             lbs = sorted(
                 (
                     lb
@@ -145,14 +156,6 @@ class RegionalELBv2View:
                 ),
                 key=lambda x: x.load_balancer_arn,
             )
-            if names:
-                found = [lb for lb in lbs if lb.load_balancer_name in names]
-                missing = set(names) - {lb.load_balancer_name for lb in found}
-                if missing:
-                    raise awserr.LoadBalancerNotFoundException(
-                        f"Load balancers '[{', '.join(sorted(missing))}]' not found"
-                    )
-                return [_copy(lb) for lb in found], None
             page, next_marker = _paginate(lbs, page_size, marker)
             return [_copy(lb) for lb in page], next_marker
 
@@ -167,8 +170,10 @@ class FakeGlobalAccelerator:
         self._tags: Dict[str, Dict[str, str]] = {}
         self._listeners: Dict[str, t.Listener] = {}  # listener arn -> listener
         self._listener_owner: Dict[str, str] = {}  # listener arn -> accelerator arn
+        self._listeners_by_acc: Dict[str, List[str]] = {}  # acc arn -> [listener arn]
         self._endpoint_groups: Dict[str, t.EndpointGroup] = {}
         self._eg_owner: Dict[str, str] = {}  # endpoint group arn -> listener arn
+        self._egs_by_listener: Dict[str, List[str]] = {}  # listener arn -> [eg arn]
 
     def _count(self, op: str):
         self.call_counts[op] = self.call_counts.get(op, 0) + 1
@@ -253,8 +258,7 @@ class FakeGlobalAccelerator:
                 raise awserr.AcceleratorNotDisabledException(
                     f"accelerator {arn} must be disabled and deployed before deletion"
                 )
-            owned = [l for l, a in self._listener_owner.items() if a == arn]
-            if owned:
+            if self._listeners_by_acc.get(arn):
                 raise awserr.AWSAPIError(
                     f"accelerator {arn} still has listeners", "AssociatedListenerFoundException"
                 )
@@ -297,6 +301,7 @@ class FakeGlobalAccelerator:
             )
             self._listeners[arn] = listener
             self._listener_owner[arn] = accelerator_arn
+            self._listeners_by_acc.setdefault(accelerator_arn, []).append(arn)
             self._mutated(accelerator_arn)
             return _copy(listener)
 
@@ -305,13 +310,7 @@ class FakeGlobalAccelerator:
             self._count('list_listeners')
             if accelerator_arn not in self._accelerators:
                 raise awserr.AcceleratorNotFoundException(accelerator_arn)
-            items = sorted(
-                (
-                    l
-                    for l, a in self._listener_owner.items()
-                    if a == accelerator_arn
-                ),
-            )
+            items = sorted(self._listeners_by_acc.get(accelerator_arn, []))
             page, token = _paginate(items, max_results, next_token)
             return [_copy(self._listeners[l]) for l in page], token
 
@@ -341,14 +340,16 @@ class FakeGlobalAccelerator:
             self._count('delete_listener')
             if listener_arn not in self._listeners:
                 raise awserr.ListenerNotFoundException(listener_arn)
-            owned = [e for e, l in self._eg_owner.items() if l == listener_arn]
-            if owned:
+            if self._egs_by_listener.get(listener_arn):
                 raise awserr.AWSAPIError(
                     f"listener {listener_arn} still has endpoint groups",
                     "AssociatedEndpointGroupFoundException",
                 )
             acc_arn = self._listener_owner.pop(listener_arn)
             del self._listeners[listener_arn]
+            owned_list = self._listeners_by_acc.get(acc_arn)
+            if owned_list and listener_arn in owned_list:
+                owned_list.remove(listener_arn)
             if acc_arn in self._accelerators:
                 self._mutated(acc_arn)
 
@@ -373,6 +374,7 @@ class FakeGlobalAccelerator:
             )
             self._endpoint_groups[arn] = eg
             self._eg_owner[arn] = listener_arn
+            self._egs_by_listener.setdefault(listener_arn, []).append(arn)
             self._mutated(self._listener_owner[listener_arn])
             return _copy(eg)
 
@@ -381,7 +383,7 @@ class FakeGlobalAccelerator:
             self._count('list_endpoint_groups')
             if listener_arn not in self._listeners:
                 raise awserr.ListenerNotFoundException(listener_arn)
-            items = sorted(e for e, l in self._eg_owner.items() if l == listener_arn)
+            items = sorted(self._egs_by_listener.get(listener_arn, []))
             page, token = _paginate(items, max_results, next_token)
             return [_copy(self._endpoint_groups[e]) for e in page], token
 
@@ -461,6 +463,9 @@ class FakeGlobalAccelerator:
                 raise awserr.EndpointGroupNotFoundException(endpoint_group_arn)
             listener_arn = self._eg_owner.pop(endpoint_group_arn)
             del self._endpoint_groups[endpoint_group_arn]
+            owned_list = self._egs_by_listener.get(listener_arn)
+            if owned_list and endpoint_group_arn in owned_list:
+                owned_list.remove(endpoint_group_arn)
             acc_arn = self._listener_owner.get(listener_arn)
             if acc_arn:
                 self._mutated(acc_arn)

@@ -63,13 +63,15 @@ def hostname_annotation_changed(old, new) -> bool:
 def objects_equal(old, new) -> bool:
     """reflect.DeepEqual guard used by every update notification.
 
-    Informer-delivered pairs are equal exactly when they carry the same
-    resourceVersion (a resync re-delivers the identical cached object —
-    which client-go's DeepEqual also skips, since UpdateFunc gets the same
-    pointer twice).  O(1) instead of serializing both objects."""
+    resourceVersion is a field of ObjectMeta, so Go's reflect.DeepEqual
+    is false whenever the rvs differ and true for a resync's identical
+    pair — i.e. for informer-delivered objects the whole comparison
+    reduces to the rv, O(1).  Content comparison only remains for
+    hand-built objects without rvs (unit tests)."""
     old_rv = old.metadata.resource_version
-    if old_rv and old_rv == new.metadata.resource_version:
-        return True
+    new_rv = new.metadata.resource_version
+    if old_rv or new_rv:
+        return old_rv == new_rv
     return to_dict(old) == to_dict(new)
 
 

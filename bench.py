@@ -130,9 +130,7 @@ def converged(backend, owner_to_port: dict) -> bool:
     the convergence poll doesn't stall reconcile workers."""
     ga = backend.ga
     with backend.lock:
-        listeners_by_acc = {}
-        for listener_arn, acc_arn in ga._listener_owner.items():
-            listeners_by_acc.setdefault(acc_arn, []).append(listener_arn)
+        listeners_by_acc = ga._listeners_by_acc
         seen = 0
         for arn, tags in ga._tags.items():
             want = owner_to_port.get(tags.get("aws-global-accelerator-owner"))
@@ -179,13 +177,16 @@ def run_step(client, backend, services, step_idx: int, timeout: float = 120.0,
         binding.spec.weight = weight
         client.update(binding)
     deadline = time.monotonic() + timeout
+    # adaptive poll: the convergence check is O(objects) under the backend
+    # lock, so polling every 1ms would contend with the workers at scale
+    poll = max(0.001, len(services) / 64_000)
     while not (
         converged(backend, owner_to_port)
         and (not bindings or bindings_converged(client, backend, bindings, weight))
     ):
         if time.monotonic() >= deadline:
             raise TimeoutError(f"step {step_idx} did not converge in {timeout}s")
-        time.sleep(0.001)
+        time.sleep(poll)
 
 
 def main():
@@ -223,10 +224,11 @@ def main():
         # initial creation converges during warmup setup
         owner_to_port = {f"service/default/{n}": 80 for n in services}
         deadline = time.monotonic() + 120.0
+        poll = max(0.001, args.objects / 64_000)
         while not converged(backend, owner_to_port):
             if time.monotonic() >= deadline:
                 raise TimeoutError("initial convergence timed out")
-            time.sleep(0.001)
+            time.sleep(poll)
 
         for w in range(args.warmup):
             run_step(client, backend, services, w, bindings=bindings)
