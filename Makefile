@@ -9,10 +9,12 @@ test:
 test-fast:
 	$(PYTHON) -m pytest tests/ -q -m "not gpu" -n 4
 
-# threading stress profile: run the suite repeatedly with randomized order
+# threading stress profile: repeat the concurrency-heavy suites
 test-race:
-	$(PYTHON) -m pytest tests/ -q -m "not gpu" -p no:cacheprovider --count 3 2>/dev/null \
-		|| $(PYTHON) -m pytest tests/ -q -m "not gpu"
+	for i in 1 2 3; do \
+		$(PYTHON) -m pytest tests/test_stress_concurrency.py tests/test_chaos_recovery.py \
+			tests/test_store_event_sourcing.py -q || exit 1; \
+	done
 
 bench:
 	$(PYTHON) bench.py --steps 10 --warmup 3

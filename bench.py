@@ -179,7 +179,7 @@ def run_step(client, backend, services, step_idx: int, timeout: float = 120.0,
     deadline = time.monotonic() + timeout
     # adaptive poll: the convergence check is O(objects) under the backend
     # lock, so polling every 1ms would contend with the workers at scale
-    poll = max(0.001, len(services) / 64_000)
+    poll = max(0.0003, len(services) / 64_000)
     while not (
         converged(backend, owner_to_port)
         and (not bindings or bindings_converged(client, backend, bindings, weight))
@@ -224,7 +224,7 @@ def main():
         # initial creation converges during warmup setup
         owner_to_port = {f"service/default/{n}": 80 for n in services}
         deadline = time.monotonic() + 120.0
-        poll = max(0.001, args.objects / 64_000)
+        poll = max(0.0003, args.objects / 64_000)
         while not converged(backend, owner_to_port):
             if time.monotonic() >= deadline:
                 raise TimeoutError("initial convergence timed out")
