@@ -288,3 +288,42 @@ class TestK8sClient:
             assert wait_until(lambda: len(backend.ga.list_accelerators()[0]) == 1)
         finally:
             stop.set()
+
+
+class TestInClusterConfig:
+    def test_reads_service_account_mount(self, tmp_path, monkeypatch):
+        from agac.kube import kubeconfig as kc
+
+        sa = tmp_path / "serviceaccount"
+        sa.mkdir()
+        (sa / "token").write_text("sa-token\n")
+        (sa / "ca.crt").write_text("CA PEM")
+        monkeypatch.setattr(kc, "SERVICE_ACCOUNT_DIR", str(sa))
+        monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+        monkeypatch.setenv("KUBERNETES_SERVICE_PORT", "6443")
+        rest = kc.in_cluster_config()
+        assert rest.host == "https://10.0.0.1:6443"
+        assert rest.token == "sa-token"
+        assert rest.ca_cert == str(sa / "ca.crt")
+
+    def test_outside_cluster_raises(self, monkeypatch):
+        from agac.kube import kubeconfig as kc
+
+        monkeypatch.delenv("KUBERNETES_SERVICE_HOST", raising=False)
+        with pytest.raises(RuntimeError):
+            kc.in_cluster_config()
+
+
+class TestClientAuthWiring:
+    def test_bearer_token_header(self, api):
+        client = K8sKubeClient(RestConfig(host=api.url, token="tok-123"))
+        assert client.session.headers["Authorization"] == "Bearer tok-123"
+        # and requests still work (the test server ignores auth)
+        items, _ = client.list("Service")
+        assert items == []
+
+    def test_insecure_skip_verify(self, api):
+        client = K8sKubeClient(
+            RestConfig(host=api.url, insecure_skip_tls_verify=True)
+        )
+        assert client.session.verify is False
