@@ -118,7 +118,7 @@ class GlobalAcceleratorController:
                 self._enqueue_ingress(new)
 
     def _delete_ingress_notification(self, ingress):
-        # reference enqueues ingress deletes unconditionally (controller.go:170)
+        # reference enqueues ingress deletes unconditionally (ga/controller.go:170)
         self._enqueue_ingress(ingress)
 
     def _enqueue_service(self, obj):

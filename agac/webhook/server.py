@@ -51,7 +51,7 @@ class WebhookHandler(BaseHTTPRequestHandler):
 
     def _parse_request(self) -> Tuple[Optional[dict], Optional[str]]:
         """Content-type + AdmissionReview decoding
-        (reference webhook.go:61-85)."""
+        (reference pkg/webhoook/webhook.go:61-85)."""
         if self.headers.get("Content-Type") != "application/json":
             return None, "invalid Content-Type"
         length = int(self.headers.get("Content-Length") or 0)
