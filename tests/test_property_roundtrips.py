@@ -107,9 +107,18 @@ def test_deep_copy_is_equal_and_independent(svc):
     assert "mutated" not in svc.metadata.annotations
 
 
+# NB: an LB literally named "internal" (or "internal-…") is unparseable by
+# the reference's regexes too (load_balancer.go:50-66 strips the prefix and
+# then finds no name) — bug-compat kept, so the generator excludes it.
 lb_names = st.text(
     alphabet=string.ascii_lowercase + string.digits + "-", min_size=1, max_size=28
-).filter(lambda s: not s.startswith("-") and not s.endswith("-") and "--" not in s)
+).filter(
+    lambda s: not s.startswith("-")
+    and not s.endswith("-")
+    and "--" not in s
+    and s != "internal"
+    and not s.startswith("internal-")
+)
 hashes = st.text(alphabet=string.ascii_lowercase + string.digits, min_size=8, max_size=16)
 regions = st.sampled_from(["us-east-1", "us-west-2", "eu-central-1", "ap-northeast-1"])
 
