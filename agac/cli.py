@@ -288,6 +288,23 @@ def get_cmd(kind, namespace, master):
         )
 
 
+@cli.command("apply")
+@click.option("-f", "--filename", multiple=True, required=True, type=click.Path(exists=True), help="YAML file(s) to apply (repeatable).")
+@click.option("--master", default="", help="agac API server URL (native wire scheme).")
+def apply_cmd(filename, master):
+    """Apply YAML manifests to an agac API server (kubectl-apply-lite;
+    unknown kinds are skipped)."""
+    from .kube.apply import apply_yaml
+    from .kube.rest import RestKubeClient
+
+    server = master or os.environ.get("AGAC_API_SERVER", "http://127.0.0.1:8001")
+    client = RestKubeClient(server)
+    for path in filename:
+        with open(path) as f:
+            for action, ident in apply_yaml(client, f.read()):
+                click.echo(f"{ident} {action}")
+
+
 @cli.command()
 @click.option("--objects", default=3, show_default=True, help="Sample services to reconcile.")
 def demo(objects):
