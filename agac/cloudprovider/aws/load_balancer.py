@@ -69,9 +69,9 @@ class LoadBalancerMixin:
 
         Raises LoadBalancerNotFound (typed, from the API) or ValueError if
         the response somehow lacks the requested name."""
-        from . import metrics_shim as _m
+        from ...metrics import observe_aws_call
 
-        _m.count("elbv2", "DescribeLoadBalancers")
+        observe_aws_call("elbv2", "DescribeLoadBalancers")
         lbs, _ = self.lb.describe_load_balancers(names=[name])
         for lb in lbs:
             if lb.load_balancer_name == name:

@@ -119,6 +119,11 @@ class RateLimitingQueue:
         self._seq = itertools.count()
         self._waiting_thread: Optional[threading.Thread] = None
 
+    def _report_depth(self):
+        from .. import metrics
+
+        metrics.set_queue_depth(self.name, len(self._queue))
+
     # -- base queue --------------------------------------------------------
     def add(self, item: Any):
         with self._cond:
@@ -130,6 +135,7 @@ class RateLimitingQueue:
             if item in self._processing:
                 return
             self._queue.append(item)
+            self._report_depth()
             self._cond.notify_all()
 
     def get(self, timeout: Optional[float] = None) -> Tuple[Any, bool]:
@@ -150,6 +156,7 @@ class RateLimitingQueue:
             item = self._queue.pop(0)
             self._processing.add(item)
             self._dirty.discard(item)
+            self._report_depth()
             return item, False
 
     def done(self, item: Any):

@@ -32,6 +32,13 @@ if _AVAILABLE:
         "AWS API operations issued",
         ["service", "operation"],
     )
+    from prometheus_client import Gauge
+
+    WORKQUEUE_DEPTH = Gauge(
+        "agac_workqueue_depth",
+        "Items currently queued (not yet picked up by a worker)",
+        ["queue"],
+    )
 
 
 def observe_reconcile(queue_name: str, outcome: str, seconds: float):
@@ -43,6 +50,11 @@ def observe_reconcile(queue_name: str, outcome: str, seconds: float):
 def observe_aws_call(service: str, operation: str):
     if _AVAILABLE:
         AWS_API_CALLS.labels(service=service, operation=operation).inc()
+
+
+def set_queue_depth(queue_name: str, depth: int):
+    if _AVAILABLE and queue_name:
+        WORKQUEUE_DEPTH.labels(queue=queue_name).set(depth)
 
 
 def start_metrics_server(port: int):

@@ -24,6 +24,7 @@ import time
 from dataclasses import dataclass
 
 from . import metrics
+from .apis.meta import deep_copy
 from .errors import is_no_retry
 from .kube.store import is_not_found
 
@@ -72,8 +73,6 @@ def _reconcile_handler(key, queue, key_to_obj, process_delete, process_create_or
                 outcome = "error"
                 return
         else:
-            from .apis.meta import deep_copy
-
             res, err = _run(process_create_or_update, deep_copy(obj))
 
         if err is not None:

@@ -73,3 +73,19 @@ def test_metrics_counters_observable():
         {"service": "elbv2-test", "operation": "DescribeLoadBalancers"},
     )
     assert value == 1.0
+
+
+def test_workqueue_depth_gauge():
+    from prometheus_client import REGISTRY
+
+    from agac.kube.workqueue import ItemExponentialFailureRateLimiter, RateLimitingQueue
+
+    q = RateLimitingQueue(
+        rate_limiter=ItemExponentialFailureRateLimiter(0.001, 0.01),
+        name="depth-test-q",
+    )
+    q.add("a")
+    q.add("b")
+    assert REGISTRY.get_sample_value("agac_workqueue_depth", {"queue": "depth-test-q"}) == 2.0
+    q.get()
+    assert REGISTRY.get_sample_value("agac_workqueue_depth", {"queue": "depth-test-q"}) == 1.0
