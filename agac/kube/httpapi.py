@@ -66,6 +66,7 @@ def _error_body(e: APIError) -> dict:
 
 class _Handler(BaseHTTPRequestHandler):
     protocol_version = "HTTP/1.1"
+    disable_nagle_algorithm = True  # small JSON bodies; latency over batching
     store: APIStore = None  # set by server factory
 
     def log_message(self, fmt, *args):  # noqa: A003

@@ -20,6 +20,7 @@ logger = logging.getLogger(__name__)
 
 
 class WebhookHandler(BaseHTTPRequestHandler):
+    disable_nagle_algorithm = True
     # quiet default request logging into our logger
     def log_message(self, fmt, *args):  # noqa: A003
         logger.debug(fmt, *args)

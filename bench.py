@@ -35,7 +35,7 @@ import threading
 import time
 
 
-def build_stack(objects: int, workers: int, scenario: str = "ga"):
+def build_stack(objects: int, workers: int, scenario: str = "ga", api: str = "memory"):
     from agac.apis import core as corev1
     from agac.apis import endpointgroupbinding as egb
     from agac.apis.meta import ObjectMeta
@@ -55,7 +55,17 @@ def build_stack(objects: int, workers: int, scenario: str = "ga"):
     LB_TYPE = "service.beta.kubernetes.io/aws-load-balancer-type"
     region = "us-east-1"
 
-    client = InMemoryKubeClient()
+    if api == "http":
+        # full network boundary: in-process HTTP apiserver + REST client
+        from agac.kube.httpapi import APIServer
+        from agac.kube.rest import RestKubeClient
+        from agac.kube.store import APIStore
+
+        server = APIServer(APIStore())
+        server.start()
+        client = RestKubeClient(server.url)
+    else:
+        client = InMemoryKubeClient()
     backend = FakeAWSBackend(deploy_after_describes=0)
     factory = FakeCloudFactory(backend)
     stop = threading.Event()

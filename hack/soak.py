@@ -35,11 +35,12 @@ def main():
     parser.add_argument("--minutes", type=float, default=10.0)
     parser.add_argument("--objects", type=int, default=64)
     parser.add_argument("--scenario", choices=["ga", "full"], default="full")
+    parser.add_argument("--api", choices=["memory", "http"], default="memory")
     parser.add_argument("--report-every", type=float, default=30.0)
     args = parser.parse_args()
 
     client, backend, services, bindings, stop = bench.build_stack(
-        args.objects, workers=1, scenario=args.scenario
+        args.objects, workers=1, scenario=args.scenario, api=args.api
     )
     samples = []
     try:
@@ -74,6 +75,7 @@ def main():
         final_rss = current_rss_mb()
         growth = final_rss - (base_rss or final_rss)
         verdict = {
+            "api": args.api,
             "soak_minutes": args.minutes,
             "total_steps": step,
             "objects": args.objects + len(bindings),
