@@ -210,6 +210,11 @@ def main():
     # 8w=234/s on this container)
     parser.add_argument("--workers", type=int, default=1)
     parser.add_argument(
+        "--api", choices=["memory", "http"], default="memory",
+        help="kube API backend: in-process store (default) or an in-process "
+             "HTTP apiserver + REST client (full network boundary)",
+    )
+    parser.add_argument(
         "--scenario", choices=["ga", "full"], default="ga",
         help="ga: GlobalAccelerator service churn (the headline metric); "
              "full: adds Route53 records + EndpointGroupBinding weight churn",
@@ -227,7 +232,7 @@ def main():
         dist.init_process_group(backend="gloo")
 
     client, backend, services, bindings, stop = build_stack(
-        args.objects, args.workers, args.scenario
+        args.objects, args.workers, args.scenario, api=args.api
     )
 
     try:
@@ -292,6 +297,7 @@ def main():
                         "config": {
                             "model": "k8s-controller reconcile loop (BASELINE.json: tier-mismatch, no ML model; proxy metric = reconcile latency event->converged)",
                             "objects_per_rank": args.objects,
+                            "api": args.api,
                             "scenario": args.scenario,
                             "bindings_per_rank": len(bindings),
                             "workers_per_queue": args.workers,

@@ -67,3 +67,17 @@ def test_bench_world_size_2_gloo():
     # whole-job aggregate: 8 objects × 2 steps × 2 ranks / elapsed
     assert payload["value"] > 0
     assert "x2" in payload["config"]["parallelism"]
+
+
+def test_bench_http_boundary_mode():
+    """--api http runs the same workload through the HTTP apiserver +
+    REST client (smaller config: each op is a real network round trip)."""
+    result = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--objects", "4", "--api", "http"],
+        cwd=REPO_ROOT, capture_output=True, text=True, timeout=300,
+    )
+    assert result.returncode == 0, result.stderr[-2000:]
+    payload = last_json_line(result.stdout)
+    assert payload["config"]["api"] == "http"
+    assert payload["value"] > 0
