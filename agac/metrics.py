@@ -9,7 +9,7 @@ are no-ops if prometheus_client is unavailable.
 from __future__ import annotations
 
 try:
-    from prometheus_client import Counter, Histogram, start_http_server
+    from prometheus_client import Counter, Gauge, Histogram, start_http_server
 
     _AVAILABLE = True
 except Exception:  # pragma: no cover - prometheus_client is installed here
@@ -32,8 +32,6 @@ if _AVAILABLE:
         "AWS API operations issued",
         ["service", "operation"],
     )
-    from prometheus_client import Gauge
-
     WORKQUEUE_DEPTH = Gauge(
         "agac_workqueue_depth",
         "Items currently queued (not yet picked up by a worker)",
