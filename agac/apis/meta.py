@@ -91,8 +91,22 @@ def from_dict(cls, data: dict):
 
 
 @dataclass(slots=True)
+class OwnerReference:
+    """metav1.OwnerReference — modeled so controller writes against a real
+    apiserver (full-replacement PUTs) never drop a user's owner links."""
+
+    api_version: str = ""
+    kind: str = ""
+    name: str = ""
+    uid: str = ""
+    controller: typing.Optional[bool] = None
+    block_owner_deletion: typing.Optional[bool] = None
+
+
+@dataclass(slots=True)
 class ObjectMeta:
-    """Reference: metav1.ObjectMeta (the subset the controllers use)."""
+    """Reference: metav1.ObjectMeta (the subset the controllers use, plus
+    the write-safety fields for full-replacement updates)."""
 
     name: str = ""
     namespace: str = ""
@@ -104,6 +118,7 @@ class ObjectMeta:
     annotations: typing.Dict[str, str] = field(default_factory=dict)
     labels: typing.Dict[str, str] = field(default_factory=dict)
     finalizers: typing.List[str] = field(default_factory=list)
+    owner_references: typing.List[OwnerReference] = field(default_factory=list)
 
 
 @dataclass(slots=True)

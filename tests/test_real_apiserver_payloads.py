@@ -199,3 +199,40 @@ def test_egb_from_kubectl_apply_shape():
     binding = from_dict(egb.EndpointGroupBinding, doc)
     assert binding.spec.weight == 100
     assert binding.spec.service_ref.name == "echoserver"
+
+
+def test_owner_references_survive_update_roundtrip():
+    """A real-cluster EGB with ownerReferences must not lose them when the
+    controller PUTs back finalizer/spec changes (full-replacement update)."""
+    doc = {
+        "apiVersion": "operator.h3poteto.dev/v1alpha1",
+        "kind": "EndpointGroupBinding",
+        "metadata": {
+            "name": "owned",
+            "namespace": "default",
+            "ownerReferences": [
+                {
+                    "apiVersion": "apps/v1",
+                    "kind": "Deployment",
+                    "name": "parent",
+                    "uid": "u-1",
+                    "controller": True,
+                    "blockOwnerDeletion": True,
+                }
+            ],
+        },
+        "spec": {"endpointGroupArn": "arn:x"},
+    }
+    binding = from_dict(egb.EndpointGroupBinding, doc)
+    assert binding.metadata.owner_references[0].name == "parent"
+    assert binding.metadata.owner_references[0].controller is True
+    out = to_dict(binding)
+    ref = out["metadata"]["ownerReferences"][0]
+    assert ref == {
+        "apiVersion": "apps/v1",
+        "kind": "Deployment",
+        "name": "parent",
+        "uid": "u-1",
+        "controller": True,
+        "blockOwnerDeletion": True,
+    }
