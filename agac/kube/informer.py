@@ -136,6 +136,9 @@ class Informer:
         while not stop.is_set():
             try:
                 self._list_and_watch(stop)
+            except GoneError:
+                # resourceVersion expired (apiserver 410) — relist quietly
+                logger.info("informer %s watch expired; relisting", self.kind)
             except Exception:
                 logger.exception("informer %s list/watch failed; backing off", self.kind)
                 stop.wait(1.0)
