@@ -50,9 +50,12 @@ def main():
         interval_steps = 0
         interval_start = time.monotonic()
         base_rss = None
+        latencies = []
         while time.monotonic() < deadline:
+            t0 = time.monotonic()
             bench.run_step(client, backend, services, step, timeout=120.0,
                            bindings=bindings)
+            latencies.append(time.monotonic() - t0)
             step += 1
             interval_steps += 1
             now = time.monotonic()
@@ -74,7 +77,14 @@ def main():
                 interval_start = now
         final_rss = current_rss_mb()
         growth = final_rss - (base_rss or final_rss)
+        latencies.sort()
+
+        def pct(p):
+            return round(latencies[min(len(latencies) - 1, int(p * len(latencies)))] * 1000, 2)
+
         verdict = {
+            "latency_ms": {"p50": pct(0.50), "p90": pct(0.90),
+                           "p99": pct(0.99), "max": round(latencies[-1] * 1000, 2)},
             "api": args.api,
             "soak_minutes": args.minutes,
             "total_steps": step,
