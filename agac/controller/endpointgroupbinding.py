@@ -10,9 +10,12 @@ Deliberate fixes over the reference (documented per method):
 - the delete path removes ALL endpoint ids (the reference iterates the
   slice while truncating it, ``egb/reconcile.go:70-85``, which skips every
   other element and needs extra requeue rounds);
-- endpoint operations use a per-endpoint regional client instead of
-  whatever regional client the hostname loop happened to end on
-  (``egb/reconcile.go:121-133`` reuses the last ``regionalCloud``);
+- Global Accelerator endpoint operations go through the us-west-2-homed
+  client (GA is a global service; the reference's per-region ``NewAWS``
+  pins its ga client there anyway), while the add path keeps a regional
+  client for the LB lookup it needs — replacing the reference's reuse of
+  whatever ``regionalCloud`` the hostname loop ended on
+  (``egb/reconcile.go:121-133``);
 - a failed reconcile is requeued rate-limited (the reference's
   ``processNextWorkItem`` only logs and waits for the next 30s resync,
   ``egb/controller.go:136-143``).
@@ -27,7 +30,7 @@ from dataclasses import dataclass
 from .. import reconcile
 from ..apis.endpointgroupbinding import FINALIZER
 from ..apis.meta import deep_copy, meta_namespace_key, split_meta_namespace_key
-from ..cloudprovider.aws import get_lb_name_from_hostname, get_region_from_arn
+from ..cloudprovider.aws import get_lb_name_from_hostname
 from ..cloudprovider.aws.errors import ERR_ENDPOINT_GROUP_NOT_FOUND, error_code
 from ..kube.events import EventRecorder
 from ..kube.informer import wait_for_cache_sync
@@ -197,9 +200,10 @@ class EndpointGroupBindingController:
 
         remaining = list(binding.status.endpoint_ids)
         for endpoint_id in binding.status.endpoint_ids:
-            region = get_region_from_arn(endpoint_id)
-            regional_cloud = self.cloud_factory(region)
-            regional_cloud.remove_lb_from_endpoint_group(endpoint_group, endpoint_id)
+            # GA endpoint ops go through the us-west-2-homed client: the
+            # endpoint's ARN region is irrelevant (GA is a global service,
+            # and the reference's per-region NewAWS also pins ga there)
+            cloud.remove_lb_from_endpoint_group(endpoint_group, endpoint_id)
             remaining.remove(endpoint_id)
 
         copied = deep_copy(binding)
@@ -237,8 +241,7 @@ class EndpointGroupBindingController:
 
         results = list(binding.status.endpoint_ids)
         for endpoint_id in removed_endpoint_ids:
-            regional_cloud = self.cloud_factory(get_region_from_arn(endpoint_id))
-            regional_cloud.remove_lb_from_endpoint_group(endpoint_group, endpoint_id)
+            cloud.remove_lb_from_endpoint_group(endpoint_group, endpoint_id)
             results = [e for e in results if e != endpoint_id]
 
         for endpoint_id in new_endpoint_ids:
