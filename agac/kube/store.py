@@ -235,11 +235,12 @@ class APIStore:
         from .validation import validate_object
 
         validate_object(obj)
-        try:
-            current = self.get(kind, obj.metadata.namespace, obj.metadata.name)
-        except NotFoundError:
-            current = None
-        self._admit(kind, "UPDATE", current, obj)
+        if self.admission_webhooks:
+            try:
+                current = self.get(kind, obj.metadata.namespace, obj.metadata.name)
+            except NotFoundError:
+                current = None
+            self._admit(kind, "UPDATE", current, obj)
         with self._lock:
             key = (obj.metadata.namespace, obj.metadata.name)
             bucket = self._bucket(kind)
@@ -295,11 +296,12 @@ class APIStore:
 
     def delete(self, kind: str, namespace: str, name: str):
         """Finalizer-aware delete (kube-apiserver graceful deletion)."""
-        try:
-            current = self.get(kind, namespace, name)
-        except NotFoundError:
-            current = None
-        self._admit(kind, "DELETE", current, None)
+        if self.admission_webhooks:
+            try:
+                current = self.get(kind, namespace, name)
+            except NotFoundError:
+                current = None
+            self._admit(kind, "DELETE", current, None)
         with self._lock:
             key = (namespace, name)
             bucket = self._bucket(kind)
