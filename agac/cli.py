@@ -148,6 +148,15 @@ def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_po
         cloud_factory = FakeCloudFactory()
         logger.info("Using the in-memory AWS fake")
 
+    if api in ("http", "k8s"):
+        # fail fast with a clear message if the API server is unreachable
+        try:
+            kube_client.list("Lease", "default")
+        except Exception as e:
+            raise click.ClickException(
+                f"cannot reach the {api} API server: {e}"
+            ) from e
+
     if metrics_port:
         start_metrics_server(metrics_port)
 
