@@ -161,21 +161,27 @@ class Informer:
                 self._dispatch_delete(metalib.deep_copy(obj))
         self._synced.set()
 
-        watch = self.client.watch(self.kind, resource_version=rv)
-        self._watch = watch
-        try:
-            while not stop.is_set():
-                event = watch.get(timeout=0.2)
-                if event is None:
-                    if getattr(watch, "closed", False):
-                        return  # watch severed → relist
-                    continue
-                self._handle_event(event)
-        except GoneError:
-            return  # relist
-        finally:
-            watch.stop()
-            self._watch = None
+        # Watch loop: a closed stream re-watches from the last delivered
+        # resourceVersion (real apiservers close watches every few minutes;
+        # client-go resumes without relisting).  Only an expired rv
+        # (GoneError, apiserver 410) escapes to the caller for a relist.
+        last_rv = rv
+        while not stop.is_set():
+            watch = self.client.watch(self.kind, resource_version=last_rv)
+            self._watch = watch
+            try:
+                while not stop.is_set():
+                    event = watch.get(timeout=0.2)
+                    if event is None:
+                        if getattr(watch, "closed", False):
+                            break  # stream ended → re-watch from last_rv
+                        continue
+                    self._handle_event(event)
+                    if event.resource_version > last_rv:
+                        last_rv = event.resource_version
+            finally:
+                watch.stop()
+                self._watch = None
 
     def _handle_event(self, event):
         obj = event.obj

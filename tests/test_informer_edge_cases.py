@@ -128,3 +128,38 @@ class TestStoreFieldImmutability:
         event = watch.get(timeout=10.0)
         assert event.obj.metadata.name == "mine"
         watch.stop()
+
+
+class TestRewatchWithoutRelist:
+    def test_severed_watch_resumes_from_last_rv(self):
+        """After a watch is cut, the informer re-watches from its last seen
+        resourceVersion (no relist): gap events arrive via replay, and no
+        duplicate ADDED is dispatched for objects already in cache."""
+        client = InMemoryKubeClient()
+        factory = SharedInformerFactory(client, resync_period=0)
+        informer = factory.services()
+        events = []
+        informer.add_event_handler(
+            on_add=lambda o: events.append(("add", o.metadata.name)),
+            on_update=lambda old, new: events.append(("upd", new.metadata.name)),
+            on_delete=lambda o: events.append(("del", o.metadata.name)),
+        )
+        stop = threading.Event()
+        factory.start(stop)
+        try:
+            assert wait_for_cache_sync(stop, informer)
+            client.create(mk_service("a"))
+            assert wait_until(lambda: ("add", "a") in events)
+
+            informer._watch.stop()  # sever
+            client.create(mk_service("b"))
+            obj = client.get("Service", "default", "a")
+            obj.metadata.annotations["x"] = "y"
+            client.update(obj)
+
+            assert wait_until(lambda: ("add", "b") in events)
+            assert wait_until(lambda: ("upd", "a") in events)
+            # replay path: "a" must not be re-ADDED (cache still has it)
+            assert events.count(("add", "a")) == 1
+        finally:
+            stop.set()
