@@ -181,3 +181,76 @@ def test_fake_aws_concurrent_ensures_single_lock_consistency():
         assert len(listeners) == 1
         groups, _ = backend.ga.list_endpoint_groups(listeners[0].listener_arn)
         assert len(groups) == 1
+
+
+def test_watch_sever_chaos_reconverges():
+    """Random watch severing while churning: the resume-from-last-rv path
+    must always reconverge the cache to store state (bounded: 2 trials)."""
+    from agac.kube.informer import SharedInformerFactory, wait_for_cache_sync
+
+    for trial in range(2):
+        client = InMemoryKubeClient()
+        factory = SharedInformerFactory(client, resync_period=0)
+        informer = factory.services()
+        stop = threading.Event()
+        factory.start(stop)
+        try:
+            assert wait_for_cache_sync(stop, informer)
+            stop_chaos = threading.Event()
+
+            def churn(tid):
+                r = random.Random((trial, tid))
+                for i in range(120):
+                    name = f"s-{r.randint(0, 10)}"
+                    try:
+                        op = r.random()
+                        if op < 0.45:
+                            client.create(
+                                corev1.Service(
+                                    metadata=ObjectMeta(name=name, namespace="d")
+                                )
+                            )
+                        elif op < 0.8:
+                            o = client.get("Service", "d", name)
+                            o.metadata.annotations["i"] = str(i)
+                            client.update(o)
+                        else:
+                            client.delete("Service", "d", name)
+                    except Exception:
+                        pass
+
+            def severer():
+                r = random.Random(trial + 999)
+                while not stop_chaos.is_set():
+                    w = informer._watch
+                    if w is not None and r.random() < 0.6:
+                        try:
+                            w.stop()
+                        except Exception:
+                            pass
+                    time.sleep(0.02)
+
+            sever_thread = threading.Thread(target=severer, daemon=True)
+            sever_thread.start()
+            threads = [threading.Thread(target=churn, args=(t,)) for t in range(3)]
+            for t in threads:
+                t.start()
+            for t in threads:
+                t.join(30)
+            stop_chaos.set()
+            sever_thread.join(5)
+
+            def converged():
+                store_state = {
+                    o.metadata.name: o.metadata.resource_version
+                    for o in client.list("Service")[0]
+                }
+                cache_state = {
+                    o.metadata.name: o.metadata.resource_version
+                    for o in informer.lister().list()
+                }
+                return store_state == cache_state
+
+            assert wait_until(converged), f"trial {trial} diverged"
+        finally:
+            stop.set()
