@@ -199,7 +199,7 @@ def test_watch_sever_chaos_reconverges():
             stop_chaos = threading.Event()
 
             def churn(tid):
-                r = random.Random((trial, tid))
+                r = random.Random(trial * 100 + tid)
                 for i in range(120):
                     name = f"s-{r.randint(0, 10)}"
                     try:
