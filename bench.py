@@ -170,7 +170,7 @@ def bindings_converged(client, backend, bindings, weight: int) -> bool:
     return True
 
 
-def run_step(client, backend, services, step_idx: int, timeout: float = 120.0,
+def run_step(client, backend, services, step_idx: int, timeout: float = 300.0,
              bindings=()):
     """Mutate every service's port (and every binding's weight in the full
     scenario) and wait for full convergence."""
@@ -238,7 +238,9 @@ def main():
     try:
         # initial creation converges during warmup setup
         owner_to_port = {f"service/default/{n}": 80 for n in services}
-        deadline = time.monotonic() + 120.0
+        # creation is the reference's O(#accelerators) discovery path per
+        # object (O(N^2) total) — scale the setup budget with N
+        deadline = time.monotonic() + max(120.0, args.objects * 0.25)
         poll = max(0.0003, args.objects / 64_000)
         while not converged(backend, owner_to_port):
             if time.monotonic() >= deadline:
