@@ -140,3 +140,43 @@ def test_alb_hostname_roundtrip(name, h, region, internal):
     parsed_name, parsed_region = get_lb_name_from_hostname(hostname)
     assert parsed_name == name
     assert parsed_region == region
+
+
+@st.composite
+def ingresses(draw):
+    spec = corev1.IngressSpec(
+        ingress_class_name=draw(st.none() | st.sampled_from(["alb", "nginx"])),
+        rules=[
+            corev1.IngressRule(
+                host=draw(names),
+                http=corev1.HTTPIngressRuleValue(
+                    paths=[
+                        corev1.HTTPIngressPath(
+                            path="/",
+                            backend=corev1.IngressBackend(
+                                service=corev1.IngressServiceBackend(
+                                    name=draw(names),
+                                    port=corev1.ServiceBackendPort(
+                                        number=draw(st.integers(1, 65535))
+                                    ),
+                                )
+                            ),
+                        )
+                    ]
+                ),
+            )
+            for _ in range(draw(st.integers(0, 2)))
+        ],
+    )
+    return corev1.Ingress(
+        metadata=ObjectMeta(name=draw(names), namespace=draw(names)),
+        spec=spec,
+    )
+
+
+@settings(max_examples=50, deadline=None)
+@given(ingresses())
+def test_ingress_wire_roundtrip_stable(ingress):
+    d1 = to_dict(ingress)
+    back = from_dict(corev1.Ingress, d1)
+    assert to_dict(back) == d1

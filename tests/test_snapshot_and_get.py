@@ -136,3 +136,45 @@ class TestGetCommand:
         result = CliRunner().invoke(cli, ["get", "pods"])
         assert result.exit_code != 0
         assert "unknown kind" in result.output
+
+
+class TestGetAliases:
+    def test_all_kind_aliases_resolve(self, tmp_path):
+        from click.testing import CliRunner
+
+        from agac.cli import cli
+        from agac.kube.httpapi import APIServer
+
+        store = APIStore()
+        store.create(corev1.Service(metadata=ObjectMeta(name="s", namespace="d")))
+        api = APIServer(store)
+        api.start()
+        try:
+            for alias in ("svc", "services", "service"):
+                result = CliRunner().invoke(cli, ["get", alias, "--master", api.url])
+                assert result.exit_code == 0, result.output
+                assert "s" in result.output
+            for alias in ("ing", "egb", "leases", "events"):
+                result = CliRunner().invoke(cli, ["get", alias, "--master", api.url])
+                assert result.exit_code == 0, result.output
+        finally:
+            api.shutdown()
+
+    def test_namespace_filter(self, tmp_path):
+        from click.testing import CliRunner
+
+        from agac.cli import cli
+        from agac.kube.httpapi import APIServer
+
+        store = APIStore()
+        store.create(corev1.Service(metadata=ObjectMeta(name="a", namespace="ns1")))
+        store.create(corev1.Service(metadata=ObjectMeta(name="b", namespace="ns2")))
+        api = APIServer(store)
+        api.start()
+        try:
+            result = CliRunner().invoke(
+                cli, ["get", "svc", "-n", "ns1", "--master", api.url]
+            )
+            assert "a" in result.output and "b" not in result.output
+        finally:
+            api.shutdown()
