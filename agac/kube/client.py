@@ -57,6 +57,9 @@ class _NamespacedResource:
     def delete(self, name: str):
         return self._client.delete(self._kind, self._namespace, name)
 
+    def patch(self, name: str, patch: dict, subresource=None):
+        return self._client.patch(self._kind, self._namespace, name, patch, subresource)
+
 
 class KubeClient:
     """Abstract client: CRUD + list/watch for every kind in the scheme."""
@@ -78,6 +81,11 @@ class KubeClient:
         raise NotImplementedError
 
     def delete(self, kind: str, namespace: str, name: str):
+        raise NotImplementedError
+
+    def patch(self, kind: str, namespace: str, name: str, patch: dict,
+              subresource=None):
+        """JSON merge patch; implemented by every backend."""
         raise NotImplementedError
 
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
@@ -125,6 +133,10 @@ class InMemoryKubeClient(KubeClient):
 
     def delete(self, kind: str, namespace: str, name: str):
         return self.store.delete(kind, namespace, name)
+
+    def patch(self, kind: str, namespace: str, name: str, patch: dict,
+              subresource=None):
+        return self.store.patch(kind, namespace, name, patch, subresource)
 
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
         return self.store.watch(kind, namespace, resource_version)

@@ -153,6 +153,11 @@ class _Handler(BaseHTTPRequestHandler):
                     self._json(404, k8swire.status_for_error(NotFoundError("unknown subresource")))
                     return True
                 self._json(200, self._obj_with_kind(updated))
+            elif method == "PATCH" and name is not None:
+                updated = self.store.patch(
+                    gvr.kind, namespace or "", name, self._read_body(), subresource
+                )
+                self._json(200, self._obj_with_kind(updated))
             elif method == "DELETE" and name is not None:
                 self.store.delete(gvr.kind, namespace or "", name)
                 self._json(
@@ -223,6 +228,21 @@ class _Handler(BaseHTTPRequestHandler):
                     updated = self.store.update_status(obj)
                 else:
                     updated = self.store.update(obj)
+                self._json(200, self._obj_with_kind(updated))
+            else:
+                self._json(404, {"code": 404, "reason": "NotFound", "message": "no such route"})
+        except APIError as e:
+            self._api_error(e)
+
+    def do_PATCH(self):  # noqa: N802
+        parts, query = self._route()
+        if self._k8s_route(parts, query, "PATCH"):
+            return
+        try:
+            if len(parts) in (4, 5) and parts[0] == "apis":
+                kind, ns, name = parts[1], parts[2], parts[3]
+                subresource = parts[4] if len(parts) == 5 else None
+                updated = self.store.patch(kind, ns, name, self._read_body(), subresource)
                 self._json(200, self._obj_with_kind(updated))
             else:
                 self._json(404, {"code": 404, "reason": "NotFound", "message": "no such route"})

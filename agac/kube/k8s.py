@@ -182,6 +182,19 @@ class K8sKubeClient(KubeClient):
         _raise_for(r)
         return None
 
+    def patch(self, kind: str, namespace: str, name: str, patch: dict,
+              subresource=None):
+        from .patch import MERGE_PATCH_CONTENT_TYPE
+
+        r = self.session.patch(
+            self._url(kind, namespace or None, name, subresource),
+            json=patch,
+            headers={"Content-Type": MERGE_PATCH_CONTENT_TYPE},
+            timeout=self.timeout,
+        )
+        _raise_for(r)
+        return from_dict(class_for_kind(kind), r.json())
+
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
         params = {"watch": "true"}
         if resource_version is not None:

@@ -182,6 +182,23 @@ class RestKubeClient(KubeClient):
         _raise_for(r)
         return None
 
+    def patch(self, kind: str, namespace: str, name: str, patch: dict,
+              subresource=None):
+        from .patch import MERGE_PATCH_CONTENT_TYPE
+
+        url = f"{self.base_url}/apis/{kind}/{namespace}/{name}"
+        if subresource:
+            url += f"/{subresource}"
+        r = self.session.patch(
+            url, json=patch,
+            headers={"Content-Type": MERGE_PATCH_CONTENT_TYPE},
+            timeout=self.timeout,
+        )
+        _raise_for(r)
+        from .client import class_for_kind
+
+        return from_dict(class_for_kind(kind), r.json())
+
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
         params = {}
         if namespace:

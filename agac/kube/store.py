@@ -294,6 +294,29 @@ class APIStore:
             self._broadcast(kind, MODIFIED, stored, rv)
             return metalib.deep_copy(stored)
 
+    def patch(self, kind: str, namespace: str, name: str, patch: dict,
+              subresource: str | None = None):
+        """JSON merge patch (RFC 7386) applied to the stored object; retries
+        the optimistic-concurrency loop a few times on interleaved writes."""
+        from .client import class_for_kind
+        from .patch import json_merge_patch
+
+        cls = class_for_kind(kind)
+        for _ in range(5):
+            existing = self.get(kind, namespace, name)
+            merged_dict = json_merge_patch(metalib.to_dict(existing), patch)
+            merged = metalib.from_dict(cls, merged_dict)
+            merged.metadata.namespace = namespace
+            merged.metadata.name = name
+            merged.metadata.resource_version = existing.metadata.resource_version
+            try:
+                if subresource == "status":
+                    return self.update_status(merged)
+                return self.update(merged)
+            except ConflictError:
+                continue
+        raise ConflictError(f"{kind} {namespace}/{name}: patch retries exhausted")
+
     def delete(self, kind: str, namespace: str, name: str):
         """Finalizer-aware delete (kube-apiserver graceful deletion)."""
         if self.admission_webhooks:
