@@ -336,3 +336,26 @@ class TestEndpointGroupBindingPath:
                 return True
 
         assert wait_until(gone)
+
+
+class TestEventReasonParity:
+    def test_service_route53_event_keeps_reference_typo(self, env):
+        """r53/service.go:105 emits reason 'Route53RecourdCreated' (sic);
+        the ingress path uses the corrected spelling.  Both are observable
+        API surface and must match the reference byte-for-byte."""
+        client, backend, _ = env
+        backend.route53.create_hosted_zone("example.com")
+        svc, _ = mk_lb_service(
+            backend,
+            name="typoed",
+            annotations={MANAGED: "true", HOSTNAME_ANN: "t.example.com"},
+        )
+        client.create(svc)
+        assert wait_until(
+            lambda: any(
+                e.reason == "Route53RecourdCreated"
+                for e in client.list("Event")[0]
+            )
+        )
+        reasons = {e.reason for e in client.list("Event")[0]}
+        assert "Route53RecordCreated" not in reasons  # that's the ingress path
