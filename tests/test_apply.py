@@ -89,3 +89,29 @@ def test_applied_sample_reconciles_end_to_end():
         assert acc.name == "service-default-sample-nlb"
     finally:
         stop.set()
+
+
+def test_apply_cli_command(tmp_path):
+    from click.testing import CliRunner
+
+    from agac.cli import cli
+    from agac.kube.httpapi import APIServer
+    from agac.kube.store import APIStore
+
+    api = APIServer(APIStore())
+    api.start()
+    try:
+        manifest = tmp_path / "svc.yaml"
+        manifest.write_text(read("nlb-public-service.yaml"))
+        result = CliRunner().invoke(
+            cli, ["apply", "-f", str(manifest), "--master", api.url]
+        )
+        assert result.exit_code == 0, result.output
+        assert "service/default/sample-nlb created" in result.output
+        # idempotent second apply
+        result = CliRunner().invoke(
+            cli, ["apply", "-f", str(manifest), "--master", api.url]
+        )
+        assert "unchanged" in result.output
+    finally:
+        api.shutdown()
