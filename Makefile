@@ -1,7 +1,7 @@
 # Developer entry points (reference Makefile: build/test/codegen/manifests).
 PYTHON ?= python3
 
-.PHONY: test test-race bench run-controller run-webhook manifests-validate lint
+.PHONY: test test-fast test-race bench bench-full soak e2e run-controller run-webhook manifests-validate
 
 test:
 	$(PYTHON) -m pytest tests/ -q -m "not gpu"
@@ -18,6 +18,15 @@ test-race:
 
 bench:
 	$(PYTHON) bench.py --steps 10 --warmup 3
+
+bench-full:
+	$(PYTHON) bench.py --steps 10 --warmup 3 --scenario full
+
+soak:
+	$(PYTHON) hack/soak.py --minutes 2 --objects 64 --scenario full
+
+e2e:
+	bash hack/run-e2e.sh
 
 run-controller:
 	$(PYTHON) -m agac.cli -v controller --api memory --no-leader-elect
