@@ -12,6 +12,8 @@ from .workqueue import RateLimitingQueue, ItemExponentialFailureRateLimiter
 from .informer import SharedInformerFactory, Informer, Lister, wait_for_cache_sync
 from .leaderelection import LeaderElector, LeaderElectionConfig
 from .admission import AdmissionDeniedError, http_admission, local_admission
+from .patch import json_merge_patch
+from .apply import apply_yaml
 
 __all__ = [
     "APIStore",
@@ -32,4 +34,6 @@ __all__ = [
     "AdmissionDeniedError",
     "http_admission",
     "local_admission",
+    "json_merge_patch",
+    "apply_yaml",
 ]

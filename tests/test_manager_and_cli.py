@@ -90,3 +90,20 @@ class TestDemo:
         assert result.exit_code == 0, result.output
         assert "demo OK" in result.output
         assert "GlobalAcceleratorCreated" in result.output
+
+
+class TestLogFormat:
+    def test_json_log_format_flag(self):
+        import json as jsonlib
+        import subprocess
+        import sys
+
+        out = subprocess.run(
+            [sys.executable, "-m", "agac.cli", "--log-format", "json", "-v", "version"],
+            capture_output=True, text=True, timeout=60,
+        )
+        assert out.returncode == 0
+        for line in out.stderr.strip().splitlines():
+            if line.startswith("{"):
+                entry = jsonlib.loads(line)
+                assert {"ts", "level", "logger", "message"} <= set(entry)
