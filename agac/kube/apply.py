@@ -49,11 +49,12 @@ def apply_document(client, doc: dict) -> Tuple[str, str]:
     obj.metadata.resource_version = existing.metadata.resource_version
     from ..apis.meta import to_dict
 
-    if to_dict(obj).get("spec") == to_dict(existing).get("spec") and to_dict(
-        obj
-    ).get("metadata", {}).get("annotations") == to_dict(existing).get(
-        "metadata", {}
-    ).get("annotations"):
+    desired, live = to_dict(obj), to_dict(existing)
+    unchanged = desired.get("spec") == live.get("spec") and all(
+        desired.get("metadata", {}).get(k) == live.get("metadata", {}).get(k)
+        for k in ("annotations", "labels")
+    )
+    if unchanged:
         return "unchanged", ident
     client.update(obj)
     return "configured", ident

@@ -115,3 +115,15 @@ def test_apply_cli_command(tmp_path):
         assert "unchanged" in result.output
     finally:
         api.shutdown()
+
+
+def test_label_change_reconfigures():
+    client = InMemoryKubeClient()
+    text = read("nlb-public-service.yaml")
+    apply_yaml(client, text)
+    labeled = text.replace(
+        "metadata:\n  name: sample-nlb",
+        "metadata:\n  name: sample-nlb\n  labels:\n    team: infra",
+    )
+    assert apply_yaml(client, labeled) == [("configured", "service/default/sample-nlb")]
+    assert client.get("Service", "default", "sample-nlb").metadata.labels == {"team": "infra"}
