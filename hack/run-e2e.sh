@@ -5,17 +5,17 @@
 set -euo pipefail
 cd "$(dirname "$0")/.."
 
-python -m agac.cli apiserver --port 18001 &
+python -m agac.cli apiserver --port 18101 &
 API_PID=$!
-python -m agac.cli webhook --no-ssl --port 18443 &
+python -m agac.cli webhook --no-ssl --port 18543 &
 WEBHOOK_PID=$!
 trap 'kill $API_PID $WEBHOOK_PID 2>/dev/null || true' EXIT
 
 for i in $(seq 50); do
-  curl -fsS http://127.0.0.1:18001/healthz >/dev/null 2>&1 && break
+  curl -fsS http://127.0.0.1:18101/healthz >/dev/null 2>&1 && break
   sleep 0.2
 done
-curl -fsS http://127.0.0.1:18443/healthz
+curl -fsS http://127.0.0.1:18543/healthz
 
 python -m pytest tests/test_http_e2e.py tests/test_full_e2e_scenarios.py \
   tests/test_k8s_wire.py tests/test_cli_processes.py -q "$@"
