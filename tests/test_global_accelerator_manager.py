@@ -384,3 +384,26 @@ class TestUserTagDrift:
             svc, lb_ingress(lb), "c", "mylb", REGION
         )
         assert backend.ga.describe_accelerator(arn).name == "renamed"
+
+    def test_removed_user_tag_persists_like_reference(self, backend, cloud):
+        """Shared quirk: TagResource merges and the drift predicate only
+        checks target ⊆ actual (reference global_accelerator.go:426-436,
+        :730-738), so a user tag removed from the annotation stays on the
+        accelerator in both implementations."""
+        svc = mk_service(
+            annotations={
+                "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-tags": "env=prod"
+            }
+        )
+        lb = seed_lb(backend)
+        arn, _, _ = cloud.ensure_global_accelerator_for_service(
+            svc, lb_ingress(lb), "c", "mylb", REGION
+        )
+        del svc.metadata.annotations[
+            "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-tags"
+        ]
+        cloud.ensure_global_accelerator_for_service(
+            svc, lb_ingress(lb), "c", "mylb", REGION
+        )
+        tags = {x.key: x.value for x in backend.ga.list_tags_for_resource(arn)}
+        assert tags.get("env") == "prod"  # stale but reference-faithful
