@@ -209,3 +209,57 @@ class TestAcceleratorHelpers:
         )
         assert endpoint_contains_lb(group, lb)
         assert not endpoint_contains_lb(t.EndpointGroup(), lb)
+
+
+class TestProtocolChangeReferenceTable:
+    """The six cases of the reference's TestListenerProtocolChange table
+    (global_accelerator_test.go:15-155), 1:1 — last-typed-port-wins decides."""
+
+    def test_not_changed_single(self):
+        assert not listener_protocol_changed_from_service(
+            listener([80]), svc_with_ports((80, "TCP"))
+        )
+
+    def test_not_changed_multiple_same(self):
+        assert not listener_protocol_changed_from_service(
+            listener([80]), svc_with_ports((80, "TCP"), (443, "TCP"))
+        )
+
+    def test_not_changed_multiple_different_tcp_last(self):
+        # UDP then TCP: last port's protocol (TCP) matches the listener
+        assert not listener_protocol_changed_from_service(
+            listener([80]), svc_with_ports((53, "UDP"), (80, "TCP"))
+        )
+
+    def test_changed_single(self):
+        assert listener_protocol_changed_from_service(
+            listener([53]), svc_with_ports((53, "UDP"))
+        )
+
+    def test_changed_multiple_udp(self):
+        assert listener_protocol_changed_from_service(
+            listener([53]), svc_with_ports((53, "UDP"), (54, "UDP"))
+        )
+
+    def test_changed_multiple_different_udp_last(self):
+        # TCP then UDP: last is UDP, listener TCP → drift
+        assert listener_protocol_changed_from_service(
+            listener([80]), svc_with_ports((80, "TCP"), (53, "UDP"))
+        )
+
+
+class TestPortChangeReferenceTable:
+    """Remaining cases of TestListenerPortChanged
+    (global_accelerator_test.go:157-343)."""
+
+    def test_multiple_ports_changed(self):
+        svc = svc_with_ports((8080, "TCP"), (9090, "TCP"))
+        assert listener_port_changed_from_service(listener([80, 443]), svc)
+
+    def test_ports_increased(self):
+        svc = svc_with_ports((80, "TCP"), (443, "TCP"), (8443, "TCP"))
+        assert listener_port_changed_from_service(listener([80, 443]), svc)
+
+    def test_ports_decreased(self):
+        svc = svc_with_ports((80, "TCP"))
+        assert listener_port_changed_from_service(listener([80, 443]), svc)
