@@ -89,3 +89,36 @@ class TestDetectCloudProvider:
     def test_short_hostname(self):
         with pytest.raises(ValueError):
             detect_cloud_provider("localhost")
+
+
+class TestReferenceHostnameTable:
+    """The reference's exact 4-case table (load_balancer_test.go:15-41),
+    with its real-world hostname shapes."""
+
+    def test_public_nlb_hex_name(self):
+        name, region = get_lb_name_from_hostname(
+            "aa5849cde256f49faa7487bb433155b7-3f43353a6cb6f633.elb.ap-northeast-1.amazonaws.com"
+        )
+        assert name == "aa5849cde256f49faa7487bb433155b7"
+        assert region == "ap-northeast-1"
+
+    def test_internal_nlb(self):
+        name, region = get_lb_name_from_hostname(
+            "test-b6cdc5fbd1d6fa43.elb.ap-northeast-1.amazonaws.com"
+        )
+        assert name == "test"
+        assert region == "ap-northeast-1"
+
+    def test_public_alb_k8s_style(self):
+        name, region = get_lb_name_from_hostname(
+            "k8s-default-h3poteto-f1f41628db-201899272.ap-northeast-1.elb.amazonaws.com"
+        )
+        assert name == "k8s-default-h3poteto-f1f41628db"
+        assert region == "ap-northeast-1"
+
+    def test_internal_alb_k8s_style(self):
+        name, region = get_lb_name_from_hostname(
+            "internal-k8s-default-h3poteto-35ca57562f-777774719.ap-northeast-1.elb.amazonaws.com"
+        )
+        assert name == "k8s-default-h3poteto-35ca57562f"
+        assert region == "ap-northeast-1"
