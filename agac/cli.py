@@ -391,6 +391,41 @@ def demo(objects):
                 record.resource_records[0].value[:40] + "..."
             click.echo(f"  {record.type:<4} {record.name:<34} -> {target}")
 
+        # EndpointGroupBinding segment: bind demo-1's LB to an externally
+        # managed endpoint group
+        from .apis import endpointgroupbinding as egb
+
+        ext = backend.ga.create_accelerator("external-demo")
+        from .cloudprovider.aws import types as awstypes
+
+        ext_listener = backend.ga.create_listener(
+            ext.accelerator_arn, [awstypes.PortRange(80, 80)], "TCP"
+        )
+        ext_group = backend.ga.create_endpoint_group(
+            ext_listener.listener_arn, "us-east-1"
+        )
+        client.create(
+            egb.EndpointGroupBinding(
+                metadata=ObjectMeta(name="demo-binding", namespace="default"),
+                spec=egb.EndpointGroupBindingSpec(
+                    endpoint_group_arn=ext_group.endpoint_group_arn,
+                    weight=128,
+                    service_ref=egb.ServiceReference(name="demo-1"),
+                ),
+            )
+        )
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            binding = client.get("EndpointGroupBinding", "default", "demo-binding")
+            if binding.status.endpoint_ids:
+                break
+            time.sleep(0.05)
+        click.echo(
+            f"\nEndpointGroupBinding demo-binding: endpoints="
+            f"{len(binding.status.endpoint_ids)} weight=128 "
+            f"(finalizer={binding.metadata.finalizers})"
+        )
+
         click.echo(f"\ndeleting Service demo-0 ...")
         client.delete("Service", "default", "demo-0")
         deadline = time.monotonic() + 30
