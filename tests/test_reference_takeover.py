@@ -202,3 +202,53 @@ def test_adopts_reference_route53_records_without_churn(stack):
     assert all(
         e.reason != "Route53RecourdCreated" for e in client.list("Event")[0]
     )
+
+
+def test_adopts_reference_ingress_accelerator(stack):
+    """Ingress-owned takeover: owner tag resource prefix 'ingress/'."""
+    client, backend = stack
+    lb = backend.elbv2.create_load_balancer(
+        "legacy-ing", region=REGION, lb_type="application"
+    )
+    acc = backend.ga.create_accelerator(
+        name="ingress-default-legacy-ing",
+        tags=[
+            t.Tag("aws-global-accelerator-controller-managed", "true"),
+            t.Tag("aws-global-accelerator-owner", "ingress/default/legacy-ing"),
+            t.Tag("aws-global-accelerator-target-hostname", lb.dns_name),
+            t.Tag("aws-global-accelerator-cluster", CLUSTER),
+        ],
+    )
+    listener = backend.ga.create_listener(
+        acc.accelerator_arn, [t.PortRange(80, 80)], "TCP"
+    )
+    backend.ga.create_endpoint_group(
+        listener.listener_arn, REGION,
+        endpoint_configurations=[
+            t.EndpointConfiguration(endpoint_id=lb.load_balancer_arn)
+        ],
+    )
+    backend.ga.call_counts.clear()
+
+    client.create(
+        corev1.Ingress(
+            metadata=ObjectMeta(
+                name="legacy-ing", namespace="default",
+                annotations={
+                    MANAGED: "true",
+                    "alb.ingress.kubernetes.io/listen-ports": '[{"HTTP": 80}]',
+                },
+            ),
+            spec=corev1.IngressSpec(ingress_class_name="alb"),
+            status=corev1.IngressStatus(
+                load_balancer=corev1.IngressLoadBalancerStatus(
+                    ingress=[corev1.IngressLoadBalancerIngress(hostname=lb.dns_name)]
+                )
+            ),
+        )
+    )
+    assert wait_until(lambda: backend.ga.call_counts.get("list_accelerators", 0) >= 1)
+    time.sleep(0.3)
+    accs, _ = backend.ga.list_accelerators()
+    assert len(accs) == 1
+    assert backend.ga.call_counts.get("create_accelerator", 0) == 0
