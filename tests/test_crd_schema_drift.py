@@ -85,3 +85,24 @@ def test_k8swire_registry_matches_crd():
     assert gvr.group == crd["spec"]["group"]
     assert gvr.plural == crd["spec"]["names"]["plural"]
     assert gvr.version == crd["spec"]["versions"][0]["name"]
+
+
+def test_sample_manifest_validates_against_schema_and_store():
+    """The shipped EGB sample must satisfy both the CRD yaml constraints
+    and the store-side validation."""
+    import os
+
+    from agac.apis.meta import from_dict
+    from agac.kube.client import InMemoryKubeClient
+
+    path = os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "config", "samples", "endpointgroupbinding.yaml",
+    )
+    with open(path) as f:
+        doc = yaml.safe_load(f)
+    binding = from_dict(egb.EndpointGroupBinding, doc)
+    assert binding.spec.endpoint_group_arn.startswith("arn:aws:globalaccelerator")
+    client = InMemoryKubeClient()
+    created = client.create(binding)  # store validation passes
+    assert created.metadata.generation == 1
