@@ -127,3 +127,14 @@ class TestServerHTTP:
         with pytest.raises(urllib.error.HTTPError) as exc:
             post(server, "/nope", b"{}")
         assert exc.value.code == 404
+
+
+class TestOperationScope:
+    def test_delete_and_connect_ops_allowed(self):
+        """Reference validator.go:22-26: any non-UPDATE operation passes
+        (DELETE, CONNECT, CREATE) — the webhook only guards mutation of
+        the ARN."""
+        for op in ("DELETE", "CONNECT", "CREATE"):
+            review = admission_review(None, endpoint_group_binding(), operation=op)
+            response = validate(review)
+            assert response["response"]["allowed"] is True, op
