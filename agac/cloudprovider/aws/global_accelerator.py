@@ -261,14 +261,26 @@ class GlobalAcceleratorMixin:
         )
 
     def _verified_hint(
-        self, hint_arn: str, cluster_name: str, resource: str, ns: str, name: str
+        self,
+        hint_arn: str,
+        cluster_name: str,
+        resource: str,
+        ns: str,
+        name: str,
+        hostname: str,
     ):
-        """Fetch the hinted accelerator and verify our ownership tags;
-        returns [accelerator] or None to force the full discovery scan."""
+        """Fetch the hinted accelerator and verify our ownership tags —
+        including the target-hostname tag, so a resource with multiple LB
+        hostnames (one owned accelerator per hostname) never resolves the
+        hint to the *other* hostname's accelerator; any mismatch falls back
+        to the full ``list_global_accelerator_by_resource`` scan, which is
+        the reference's only discovery path (global_accelerator.go:87-110).
+        Returns [accelerator] or None to force the full scan."""
         try:
             accelerator = self._get_accelerator(hint_arn)
             tags = self._list_tags_for_accelerator(hint_arn)
         except Exception:
+            metrics.observe_hint("globalaccelerator", "error")
             return None
         if tags_contains_all_values(
             tags,
@@ -278,9 +290,12 @@ class GlobalAcceleratorMixin:
                     resource, ns, name
                 ),
                 GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY: cluster_name,
+                GLOBAL_ACCELERATOR_TARGET_HOSTNAME_KEY: hostname,
             },
         ):
+            metrics.observe_hint("globalaccelerator", "hit")
             return [accelerator]
+        metrics.observe_hint("globalaccelerator", "stale")
         return None
 
     def _ensure_global_accelerator(
@@ -309,7 +324,12 @@ class GlobalAcceleratorMixin:
         accelerators = None
         if hint_arn:
             accelerators = self._verified_hint(
-                hint_arn, cluster_name, resource, obj.metadata.namespace, obj.metadata.name
+                hint_arn,
+                cluster_name,
+                resource,
+                obj.metadata.namespace,
+                obj.metadata.name,
+                hostname,
             )
         if accelerators is None:
             accelerators = self.list_global_accelerator_by_resource(

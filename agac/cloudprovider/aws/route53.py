@@ -116,6 +116,7 @@ class Route53Mixin:
             accelerator = self._get_accelerator(hint_arn)
             tags = self._list_tags_for_accelerator(hint_arn)
         except Exception:
+            metrics.observe_hint("route53", "error")
             return None
         if tags_contains_all_values(
             tags,
@@ -125,7 +126,9 @@ class Route53Mixin:
                 GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY: cluster_name,
             },
         ):
+            metrics.observe_hint("route53", "hit")
             return [accelerator]
+        metrics.observe_hint("route53", "stale")
         return None
 
     def _ensure_route53(
@@ -221,14 +224,16 @@ class Route53Mixin:
                 hosted_zone.id, max_items=10, start_token=token,
                 start_record_name=hostname,
             )
-            done = False
+            matched_this_page = False
             for rs in page:
                 if rs.name == target:
                     records_at_name.append(rs)
-                elif rs.name > target:
-                    done = True
-                    break
-            if done or token is None:
+                    matched_this_page = True
+            # Route53 lists names in reversed-label DNS order, not plain
+            # lexicographic order, so the only safe early exit is "this page
+            # held no record at the target name" — StartRecordName positions
+            # the scan at the name, so all matches are in a contiguous prefix.
+            if not matched_this_page or token is None:
                 break
         owned = any(
             record.value == owner_value
@@ -241,7 +246,8 @@ class Route53Mixin:
             (
                 rs
                 for rs in records_at_name
-                if rs.alias_target is not None
+                if rs.type == t.RR_TYPE_A
+                and rs.alias_target is not None
                 and replace_wildcards(rs.name) == hostname + "."
             ),
             None,

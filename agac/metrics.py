@@ -37,12 +37,24 @@ if _AVAILABLE:
         "Items currently queued (not yet picked up by a worker)",
         ["queue"],
     )
+    HINT_TOTAL = Counter(
+        "agac_hint_total",
+        "ARN-hint verification outcomes (hit = O(1) discovery; "
+        "miss/error = fall back to the full accelerator scan)",
+        ["controller", "outcome"],
+    )
 
 
 def observe_reconcile(queue_name: str, outcome: str, seconds: float):
     if _AVAILABLE:
         RECONCILE_TOTAL.labels(queue=queue_name or "unknown", outcome=outcome).inc()
         RECONCILE_DURATION.labels(queue=queue_name or "unknown").observe(seconds)
+
+
+def observe_hint(controller: str, outcome: str):
+    """outcome: hit | stale | error (stale = tags no longer match)."""
+    if _AVAILABLE:
+        HINT_TOTAL.labels(controller=controller, outcome=outcome).inc()
 
 
 def observe_aws_call(service: str, operation: str):

@@ -165,3 +165,92 @@ def test_controller_uses_and_invalidates_hints():
             time.sleep(0.02)
     finally:
         stop.set()
+
+
+def test_hint_for_other_hostname_falls_back(env):
+    """a resource with two LB hostnames has one owned
+    accelerator per hostname; a hint pointing at the *other* hostname's
+    accelerator must be rejected (target-hostname tag mismatch) so the
+    reconcile falls back to the reference's full scan instead of retagging
+    and repointing the wrong accelerator."""
+    from agac.cloudprovider.aws import global_accelerator as ga_mod
+
+    backend, cloud = env
+    svc, lb_a, arn_a = seed(backend, cloud)
+    # second accelerator owned by the same resource but targeting hostname B
+    # (as the reference would leave behind for a 2-hostname service)
+    lb_b = backend.elbv2.create_load_balancer("otherlb", region=REGION)
+    import agac.cloudprovider.aws.types as t
+
+    acc_b = backend.ga.create_accelerator(
+        "service-default-web-b",
+        tags=[
+            t.Tag(ga_mod.GLOBAL_ACCELERATOR_MANAGED_TAG_KEY, "true"),
+            t.Tag(
+                ga_mod.GLOBAL_ACCELERATOR_OWNER_TAG_KEY,
+                ga_mod.accelerator_owner_tag_value("service", "default", "web"),
+            ),
+            t.Tag(ga_mod.GLOBAL_ACCELERATOR_TARGET_HOSTNAME_KEY, lb_b.dns_name),
+            t.Tag(ga_mod.GLOBAL_ACCELERATOR_CLUSTER_TAG_KEY, "c"),
+        ],
+    )
+    backend.ga.call_counts.clear()
+    # reconcile hostname A with a hint pointing at accelerator B
+    cloud.ensure_global_accelerator_for_service(
+        svc, corev1.LoadBalancerIngress(hostname=lb_a.dns_name), "c", "mylb", REGION,
+        hint_arn=acc_b.accelerator_arn,
+    )
+    # hint must NOT short-circuit: the full scan ran
+    assert backend.ga.call_counts.get("list_accelerators", 0) >= 1
+    # accelerator A still targets hostname A (scan path behavior preserved)
+    tags_a = {t_.key: t_.value for t_ in backend.ga.list_tags_for_resource(arn_a)}
+    assert tags_a[ga_mod.GLOBAL_ACCELERATOR_TARGET_HOSTNAME_KEY] == lb_a.dns_name
+
+
+def test_hint_outcomes_are_counted(env):
+    """hint hits/misses are observable via
+    agac_hint_total{controller,outcome}."""
+    prometheus_client = pytest.importorskip("prometheus_client")
+    from agac import metrics
+
+    def val(outcome):
+        return metrics.HINT_TOTAL.labels(
+            controller="globalaccelerator", outcome=outcome
+        )._value.get()
+
+    backend, cloud = env
+    svc, lb, arn = seed(backend, cloud)
+    before_hit, before_stale, before_err = val("hit"), val("stale"), val("error")
+    cloud.ensure_global_accelerator_for_service(
+        svc, corev1.LoadBalancerIngress(hostname=lb.dns_name), "c", "mylb", REGION,
+        hint_arn=arn,
+    )
+    assert val("hit") == before_hit + 1
+    foreign = backend.ga.create_accelerator("foreign")
+    cloud.ensure_global_accelerator_for_service(
+        svc, corev1.LoadBalancerIngress(hostname=lb.dns_name), "c", "mylb", REGION,
+        hint_arn=foreign.accelerator_arn,
+    )
+    assert val("stale") == before_stale + 1
+    cloud.ensure_global_accelerator_for_service(
+        svc, corev1.LoadBalancerIngress(hostname=lb.dns_name), "c", "mylb", REGION,
+        hint_arn="arn:aws:globalaccelerator::1:accelerator/nonexistent",
+    )
+    assert val("error") == before_err + 1
+
+
+def test_steady_state_reconcile_is_o1_aws_calls(env):
+    """a steady-state (no-drift) hinted reconcile issues a
+    constant number of AWS calls regardless of fleet size."""
+    backend, cloud = env
+    svc, lb, arn = seed(backend, cloud, n_noise=50)
+    backend.ga.call_counts.clear()
+    cloud.ensure_global_accelerator_for_service(
+        svc, corev1.LoadBalancerIngress(hostname=lb.dns_name), "c", "mylb", REGION,
+        hint_arn=arn,
+    )
+    total_ga_calls = sum(backend.ga.call_counts.values())
+    # describe_accelerator + list_tags (verify) + listener/endpoint-group
+    # reads for the drift predicates — constant, independent of the 50 noise
+    # accelerators
+    assert total_ga_calls <= 8, backend.ga.call_counts

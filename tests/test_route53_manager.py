@@ -233,3 +233,44 @@ class TestCleanup:
         cloud.cleanup_record_set(CLUSTER, "service", "default", "web")
         assert records(backend, z1) == {}
         assert records(backend, z2) == {}
+
+
+class TestOwnedRecordDiscoveryTyping:
+    """_find_owned_a_record_at must only treat type=A alias
+    records as the managed record (reference findARecord filters RRTypeA,
+    route53.go:360-367) — a pre-existing AAAA alias at the same name must
+    not be mistaken for it."""
+
+    def test_aaaa_alias_at_name_is_not_the_managed_record(self, backend, cloud):
+        zone = backend.route53.create_hosted_zone("example.com")
+        lb = backend.elbv2.create_load_balancer("mylb", region=REGION)
+        seed_ga_for_lb(backend, cloud, lb)
+        # first ensure creates TXT + A-alias
+        cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.example.com"], CLUSTER,
+        )
+        # someone adds an AAAA alias at the same name pointing elsewhere
+        backend.route53.change_resource_record_sets(zone.id, [t.Change(
+            action="CREATE",
+            record_set=t.ResourceRecordSet(
+                name="www.example.com.", type="AAAA",
+                alias_target=t.AliasTarget(
+                    hosted_zone_id="Z2BJ6XQ5FK7U4H",
+                    dns_name="stale.awsglobalaccelerator.com",
+                    evaluate_target_health=True,
+                ),
+            ),
+        )])
+        # second ensure: the managed A record is current → no upsert; the
+        # AAAA record must be left alone (not taken as the A record and
+        # "updated")
+        created, retry = cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.example.com"], CLUSTER,
+        )
+        assert not created and retry == 0
+        recs = records(backend, zone)
+        aaaa = recs[("www.example.com.", "AAAA")]
+        assert aaaa.alias_target.dns_name.rstrip(".") == "stale.awsglobalaccelerator.com"
+        assert ("www.example.com.", "A") in recs
