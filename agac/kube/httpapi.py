@@ -68,6 +68,7 @@ class _Handler(BaseHTTPRequestHandler):
     protocol_version = "HTTP/1.1"
     disable_nagle_algorithm = True  # small JSON bodies; latency over batching
     store: APIStore = None  # set by server factory
+    watch_idle_seconds = 5.0  # heartbeat/bookmark cadence on idle watches
 
     def log_message(self, fmt, *args):  # noqa: A003
         logger.debug(fmt, *args)
@@ -117,16 +118,26 @@ class _Handler(BaseHTTPRequestHandler):
                 if query.get("watch") in ("true", "1"):
                     rv = query.get("resourceVersion")
                     self._serve_watch_stream(
-                        gvr.kind, namespace, int(rv) if rv else None, k8s_style=True
+                        gvr.kind,
+                        namespace,
+                        int(rv) if rv else None,
+                        k8s_style=True,
+                        bookmarks=query.get("allowWatchBookmarks") in ("true", "1"),
                     )
                     return True
-                items, rv = self.store.list(gvr.kind, namespace)
+                limit = int(query["limit"]) if query.get("limit") else None
+                items, rv, next_token = self.store.list_page(
+                    gvr.kind, namespace, limit, query.get("continue") or None
+                )
+                metadata = {"resourceVersion": str(rv)}
+                if next_token:
+                    metadata["continue"] = next_token
                 self._json(
                     200,
                     {
                         "kind": f"{gvr.kind}List",
                         "apiVersion": gvr.api_version,
-                        "metadata": {"resourceVersion": str(rv)},
+                        "metadata": metadata,
                         "items": [self._obj_with_kind(o) for o in items],
                     },
                 )
@@ -181,11 +192,17 @@ class _Handler(BaseHTTPRequestHandler):
             if parts == ["healthz"]:
                 self._json(200, {"status": "ok"})
             elif len(parts) == 2 and parts[0] == "apis":
-                items, rv = self.store.list(parts[1], query.get("namespace"))
-                self._json(
-                    200,
-                    {"items": [self._obj_with_kind(o) for o in items], "resourceVersion": rv},
+                limit = int(query["limit"]) if query.get("limit") else None
+                items, rv, next_token = self.store.list_page(
+                    parts[1], query.get("namespace"), limit, query.get("continue") or None
                 )
+                body = {
+                    "items": [self._obj_with_kind(o) for o in items],
+                    "resourceVersion": rv,
+                }
+                if next_token:
+                    body["continue"] = next_token
+                self._json(200, body)
             elif len(parts) == 4 and parts[0] == "apis":
                 obj = self.store.get(parts[1], parts[2], parts[3])
                 self._json(200, self._obj_with_kind(obj))
@@ -268,12 +285,17 @@ class _Handler(BaseHTTPRequestHandler):
         self._serve_watch_stream(
             kind, query.get("namespace"), int(rv) if rv is not None else None,
             k8s_style=False,
+            bookmarks=query.get("allowWatchBookmarks") in ("true", "1"),
         )
 
-    def _serve_watch_stream(self, kind, namespace, rv, k8s_style: bool):
+    def _serve_watch_stream(self, kind, namespace, rv, k8s_style: bool,
+                            bookmarks: bool = False):
         """Chunked ndjson event stream.  k8s_style frames events as the real
         apiserver does ({"type", "object"}); the native scheme adds a
-        top-level resourceVersion."""
+        top-level resourceVersion.  With ``allowWatchBookmarks=true``, idle
+        periods emit BOOKMARK events carrying only metadata.resourceVersion
+        (the apiserver's watch-bookmark contract) so clients can advance
+        their resume point without object traffic."""
         watch = self.store.watch(kind, namespace, rv)
         self.send_response(200)
         self.send_header("Content-Type", "application/json")
@@ -281,10 +303,26 @@ class _Handler(BaseHTTPRequestHandler):
         self.end_headers()
         try:
             while True:
-                event = watch.get(timeout=5.0)
+                # Snapshot the rv BEFORE waiting: anything issued after this
+                # point is still in the watch queue, so bookmarking at this
+                # rv can never skip an undelivered event.
+                bookmark_rv = self.store.latest_rv()
+                event = watch.get(timeout=self.watch_idle_seconds)
                 if event is None:
-                    # heartbeat keeps half-open connections detectable
-                    self._write_chunk(b"")
+                    if bookmarks:
+                        payload = {
+                            "type": "BOOKMARK",
+                            "object": {
+                                "kind": kind,
+                                "metadata": {"resourceVersion": str(bookmark_rv)},
+                            },
+                        }
+                        if not k8s_style:
+                            payload["resourceVersion"] = bookmark_rv
+                        self._write_chunk(json.dumps(payload).encode() + b"\n")
+                    else:
+                        # heartbeat keeps half-open connections detectable
+                        self._write_chunk(b"")
                     continue
                 payload = {
                     "type": event.type,
@@ -309,8 +347,13 @@ class _Handler(BaseHTTPRequestHandler):
 class APIServer:
     """Owns the HTTP listener for one APIStore."""
 
-    def __init__(self, store: APIStore, port: int = 0, host: str = "127.0.0.1"):
-        handler = type("BoundHandler", (_Handler,), {"store": store})
+    def __init__(self, store: APIStore, port: int = 0, host: str = "127.0.0.1",
+                 watch_idle_seconds: float = 5.0):
+        handler = type(
+            "BoundHandler",
+            (_Handler,),
+            {"store": store, "watch_idle_seconds": watch_idle_seconds},
+        )
         self.httpd = ThreadingHTTPServer((host, port), handler)
         self.httpd.daemon_threads = True
         self.store = store

@@ -184,6 +184,11 @@ class Informer:
                 self._watch = None
 
     def _handle_event(self, event):
+        if event.type == "BOOKMARK":
+            # metadata-only rv advancement (apiserver watch bookmarks) —
+            # the caller already moved last_rv forward; nothing to cache
+            # or dispatch (client-go reflector parity)
+            return
         obj = event.obj
         key = (obj.metadata.namespace, obj.metadata.name)
         if event.type == "DELETED":

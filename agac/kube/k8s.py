@@ -49,12 +49,18 @@ class _K8sWatch:
                     continue
                 payload = json.loads(line)
                 obj_dict = payload["object"]
-                obj = from_dict(self._cls, obj_dict)
                 rv_raw = (obj_dict.get("metadata") or {}).get("resourceVersion", "0")
                 try:
                     rv = int(rv_raw)
                 except (TypeError, ValueError):
                     rv = 0
+                if payload["type"] == "BOOKMARK":
+                    # metadata-only frame: advances the consumer's resume rv,
+                    # never touches caches/handlers (client-go reflector
+                    # bookmark handling)
+                    self._queue.put(WatchEvent("BOOKMARK", None, rv))
+                    continue
+                obj = from_dict(self._cls, obj_dict)
                 self._queue.put(WatchEvent(payload["type"], obj, rv))
         except Exception:
             if not self._stopped:
@@ -101,16 +107,50 @@ def _raise_for(response):
 
 
 class K8sKubeClient(KubeClient):
-    def __init__(self, config: RestConfig, timeout: float = 10.0):
+    """``page_size`` chunks every LIST (``?limit=&continue=`` loop, default
+    500 — the client-go reflector's WatchListPageSize default); watches are
+    opened with ``allowWatchBookmarks=true``.  429/503 responses with
+    Retry-After are retried with backoff (apiserver priority&fairness
+    shedding) up to ``max_retries`` times."""
+
+    def __init__(self, config: RestConfig, timeout: float = 10.0,
+                 page_size: int = 500, max_retries: int = 5):
         self.config = config
         self.base_url = config.host
         self.timeout = timeout
+        self.page_size = page_size
+        self.max_retries = max_retries
         self.session = requests.Session()
         if config.token:
             self.session.headers["Authorization"] = f"Bearer {config.token}"
         if config.cert:
             self.session.cert = config.cert
         self.session.verify = config.verify
+
+    def _request(self, method: str, url: str, **kwargs):
+        """Issue a request, honoring Retry-After on 429/503 (retries never
+        apply to streaming watches — those reconnect at the informer)."""
+        import time as _time
+
+        attempt = 0
+        while True:
+            r = getattr(self.session, method)(url, **kwargs)
+            if r.status_code not in (429, 503) or attempt >= self.max_retries:
+                return r
+            retry_after = r.headers.get("Retry-After")
+            try:
+                delay = min(float(retry_after), 30.0) if retry_after else 0.0
+            except ValueError:
+                delay = 0.0
+            if not delay:
+                delay = min(0.5 * 2**attempt, 8.0)
+            logger.info(
+                "apiserver %d for %s %s; retrying in %.1fs",
+                r.status_code, method.upper(), url, delay,
+            )
+            r.close()
+            _time.sleep(delay)
+            attempt += 1
 
     # -- helpers -----------------------------------------------------------
     def _url(self, kind: str, namespace=None, name=None, subresource=None) -> str:
@@ -127,7 +167,8 @@ class K8sKubeClient(KubeClient):
     # -- verbs -------------------------------------------------------------
     def create(self, obj):
         kind = type(obj).kind
-        r = self.session.post(
+        r = self._request(
+            "post",
             self._url(kind, obj.metadata.namespace or None),
             json=self._obj_body(obj),
             timeout=self.timeout,
@@ -136,28 +177,55 @@ class K8sKubeClient(KubeClient):
         return from_dict(type(obj), r.json())
 
     def get(self, kind: str, namespace: str, name: str):
-        r = self.session.get(
-            self._url(kind, namespace or None, name), timeout=self.timeout
+        r = self._request(
+            "get", self._url(kind, namespace or None, name), timeout=self.timeout
         )
         _raise_for(r)
         return from_dict(class_for_kind(kind), r.json())
 
     def list(self, kind: str, namespace: Optional[str] = None):
-        r = self.session.get(self._url(kind, namespace), timeout=self.timeout)
-        _raise_for(r)
-        body = r.json()
+        """Paginated LIST: follows ``metadata.continue`` in ``page_size``
+        chunks; every page of one logical list is served at the first
+        page's resourceVersion.  A 410 (continue token expired) restarts
+        the whole list from scratch — the client-go reflector's
+        pagination contract (reference gets this via the informers at
+        pkg/manager/manager.go:52-53)."""
+        from .store import GoneError
+
         cls = class_for_kind(kind)
-        items = [from_dict(cls, item) for item in body.get("items", [])]
-        rv_raw = (body.get("metadata") or {}).get("resourceVersion", "0")
-        try:
-            rv = int(rv_raw)
-        except (TypeError, ValueError):
+        for attempt in (0, 1):
+            items = []
+            params = {"limit": str(self.page_size)} if self.page_size else {}
             rv = 0
-        return items, rv
+            try:
+                while True:
+                    r = self._request(
+                        "get", self._url(kind, namespace), params=params,
+                        timeout=self.timeout,
+                    )
+                    _raise_for(r)
+                    body = r.json()
+                    items.extend(from_dict(cls, item) for item in body.get("items", []))
+                    meta = body.get("metadata") or {}
+                    rv_raw = meta.get("resourceVersion", "0")
+                    try:
+                        rv = int(rv_raw)
+                    except (TypeError, ValueError):
+                        rv = 0
+                    cont = meta.get("continue")
+                    if not cont:
+                        return items, rv
+                    params = {"limit": str(self.page_size), "continue": cont}
+            except GoneError:
+                if attempt == 1:
+                    raise
+                logger.info("list %s continue token expired; restarting list", kind)
+        return items, rv  # pragma: no cover - loop always returns/raises
 
     def update(self, obj):
         kind = type(obj).kind
-        r = self.session.put(
+        r = self._request(
+            "put",
             self._url(kind, obj.metadata.namespace or None, obj.metadata.name),
             json=self._obj_body(obj),
             timeout=self.timeout,
@@ -167,7 +235,8 @@ class K8sKubeClient(KubeClient):
 
     def update_status(self, obj):
         kind = type(obj).kind
-        r = self.session.put(
+        r = self._request(
+            "put",
             self._url(kind, obj.metadata.namespace or None, obj.metadata.name, "status"),
             json=self._obj_body(obj),
             timeout=self.timeout,
@@ -176,8 +245,8 @@ class K8sKubeClient(KubeClient):
         return from_dict(type(obj), r.json())
 
     def delete(self, kind: str, namespace: str, name: str):
-        r = self.session.delete(
-            self._url(kind, namespace or None, name), timeout=self.timeout
+        r = self._request(
+            "delete", self._url(kind, namespace or None, name), timeout=self.timeout
         )
         _raise_for(r)
         return None
@@ -186,7 +255,8 @@ class K8sKubeClient(KubeClient):
               subresource=None):
         from .patch import MERGE_PATCH_CONTENT_TYPE
 
-        r = self.session.patch(
+        r = self._request(
+            "patch",
             self._url(kind, namespace or None, name, subresource),
             json=patch,
             headers={"Content-Type": MERGE_PATCH_CONTENT_TYPE},
@@ -196,7 +266,7 @@ class K8sKubeClient(KubeClient):
         return from_dict(class_for_kind(kind), r.json())
 
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
-        params = {"watch": "true"}
+        params = {"watch": "true", "allowWatchBookmarks": "true"}
         if resource_version is not None:
             params["resourceVersion"] = str(resource_version)
         r = self.session.get(

@@ -74,6 +74,11 @@ class _RestWatch:
                 if not line:
                     continue  # heartbeat
                 payload = json.loads(line)
+                if payload["type"] == "BOOKMARK":
+                    self._queue.put(
+                        WatchEvent("BOOKMARK", None, payload["resourceVersion"])
+                    )
+                    continue
                 cls = class_for_kind(payload["object"]["kind"])
                 self._queue.put(
                     WatchEvent(
@@ -145,15 +150,36 @@ class RestKubeClient(KubeClient):
         _raise_for(r)
         return from_dict(class_for_kind(kind), r.json())
 
-    def list(self, kind: str, namespace: Optional[str] = None):
-        params = {"namespace": namespace} if namespace else {}
-        r = self.session.get(
-            f"{self.base_url}/apis/{kind}", params=params, timeout=self.timeout
-        )
-        _raise_for(r)
-        body = r.json()
+    def list(self, kind: str, namespace: Optional[str] = None,
+             page_size: int = 500):
+        """Chunked list (limit/continue loop); a 410 on continuation
+        restarts the list once from scratch."""
         cls = class_for_kind(kind)
-        return [from_dict(cls, item) for item in body["items"]], body["resourceVersion"]
+        for attempt in (0, 1):
+            items, rv, cont = [], 0, None
+            try:
+                while True:
+                    params = {"limit": str(page_size)}
+                    if namespace:
+                        params["namespace"] = namespace
+                    if cont:
+                        params["continue"] = cont
+                    r = self.session.get(
+                        f"{self.base_url}/apis/{kind}", params=params,
+                        timeout=self.timeout,
+                    )
+                    _raise_for(r)
+                    body = r.json()
+                    items.extend(from_dict(cls, item) for item in body["items"])
+                    rv = body["resourceVersion"]
+                    cont = body.get("continue")
+                    if not cont:
+                        return items, rv
+            except GoneError:
+                if attempt == 1:
+                    raise
+                logger.info("list %s continue token expired; restarting list", kind)
+        return items, rv  # pragma: no cover
 
     def update(self, obj):
         meta = obj.metadata
@@ -200,7 +226,7 @@ class RestKubeClient(KubeClient):
         return from_dict(class_for_kind(kind), r.json())
 
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
-        params = {}
+        params = {"allowWatchBookmarks": "true"}
         if namespace:
             params["namespace"] = namespace
         if resource_version is not None:

@@ -22,6 +22,7 @@ modeled:
 from __future__ import annotations
 
 import itertools
+import json
 import queue
 import threading
 import time
@@ -134,6 +135,7 @@ class APIStore:
     def __init__(self):
         self._lock = threading.RLock()
         self._rv = itertools.count(1)
+        self._last_rv = 0  # highest rv issued so far (peekable, for bookmarks)
         # kind -> {(namespace, name) -> obj}
         self._objects: Dict[str, Dict[Tuple[str, str], object]] = {}
         self._watches: List[_Watch] = []
@@ -158,7 +160,15 @@ class APIStore:
         return self._objects.setdefault(kind, {})
 
     def _next_rv(self) -> int:
-        return next(self._rv)
+        self._last_rv = next(self._rv)
+        return self._last_rv
+
+    def latest_rv(self) -> int:
+        """Highest resourceVersion issued so far, without burning one.
+        Safe as a watch BOOKMARK rv: every event ≤ this rv has already been
+        broadcast (rv issue and broadcast happen under the same lock)."""
+        with self._lock:
+            return self._last_rv
 
     def _broadcast(self, kind: str, event_type: str, obj, rv: int):
         event = WatchEvent(event_type, obj, rv)
@@ -214,8 +224,67 @@ class APIStore:
                 if namespace is None or ns == namespace
             ]
             # store-level rv: highest issued so far
-            rv = next(self._rv)  # burn one to get a strictly usable marker
+            rv = self._next_rv()  # burn one to get a strictly usable marker
             return items, rv
+
+    def list_page(
+        self,
+        kind: str,
+        namespace: Optional[str] = None,
+        limit: Optional[int] = None,
+        continue_token: Optional[str] = None,
+    ):
+        """Chunked list, apiserver-style (``?limit=&continue=``).
+
+        Returns (objects, resourceVersion, next_continue_token | None).
+        Every page of one logical list carries the SAME resourceVersion
+        (the one minted for the first page, carried in the opaque token).
+        A continuation whose rv window has been evicted from the event log
+        raises ``GoneError`` (the apiserver's 410 reason=Expired), telling
+        the client to restart the list from scratch — exactly the
+        client-go reflector contract."""
+        import base64
+
+        with self._lock:
+            after: Tuple[str, str] = ("", "")
+            if continue_token:
+                try:
+                    decoded = json.loads(
+                        base64.urlsafe_b64decode(continue_token.encode()).decode()
+                    )
+                    rv = int(decoded["rv"])
+                    after = (decoded["ns"], decoded["name"])
+                except Exception:
+                    raise GoneError("malformed continue token")
+                # Expire continuations that predate the replayable window:
+                # objects created/changed since the first page would be
+                # invisibly skipped, and the client can no longer watch from
+                # the page rv either.
+                logged = [e for (k, e) in self._event_log if k == kind]
+                if (
+                    logged
+                    and logged[0].resource_version > rv + 1
+                    and len(self._event_log) == self._event_log.maxlen
+                ):
+                    raise GoneError("continue token expired")
+            else:
+                rv = self._next_rv()
+
+            keys = sorted(
+                key
+                for key in self._bucket(kind)
+                if namespace is None or key[0] == namespace
+            )
+            remaining = [k for k in keys if k > after]
+            page_keys = remaining if limit is None else remaining[:limit]
+            items = [metalib.deep_copy(self._bucket(kind)[k]) for k in page_keys]
+            next_token = None
+            if limit is not None and len(remaining) > limit:
+                ns, name = page_keys[-1]
+                next_token = base64.urlsafe_b64encode(
+                    json.dumps({"rv": rv, "ns": ns, "name": name}).encode()
+                ).decode()
+            return items, rv, next_token
 
     def _check_rv(self, existing, obj):
         if (
@@ -362,7 +431,7 @@ class APIStore:
             ]
             # peek the counter without burning (itertools.count has no peek:
             # burn one and record it as the floor for the restored counter)
-            next_rv = next(self._rv)
+            next_rv = self._next_rv()
             return {"version": 1, "nextResourceVersion": next_rv, "objects": objects}
 
     @classmethod
