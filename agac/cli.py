@@ -94,8 +94,9 @@ def resolve_kubeconfig(kubeconfig: str) -> str:
 @click.option("--api", type=click.Choice(["memory", "http", "k8s"]), default="memory", show_default=True, help="Kube API backend: embedded in-memory store, an agac HTTP API server, or a real Kubernetes API server (kubeconfig/in-cluster auth).")
 @click.option("--cloud", type=click.Choice(["auto", "aws", "fake"]), default="auto", show_default=True, help="Cloud backend: auto (aws for http/k8s APIs, fake for memory), aws (boto3, required), fake (in-memory).")
 @click.option("--metrics-port", default=0, help="Serve Prometheus metrics on this port (0 = disabled).")
+@click.option("--cloud-resync-minutes", default=0.0, show_default=True, help="Re-enqueue ALL managed objects every N minutes even if unchanged, repairing cloud-side drift (0 = disabled, matching the reference: drift on unchanged objects is never repaired — docs/PARITY.md).")
 @click.option("--leader-elect/--no-leader-elect", default=True, show_default=True)
-def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_port, leader_elect):
+def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_port, cloud_resync_minutes, leader_elect):
     """Start controller."""
     from .controller.endpointgroupbinding import EndpointGroupBindingConfig
     from .controller.globalaccelerator import GlobalAcceleratorConfig
@@ -161,10 +162,19 @@ def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_po
         start_metrics_server(metrics_port)
 
     namespace = os.environ.get("POD_NAMESPACE", "default")
+    cloud_resync = max(0.0, cloud_resync_minutes) * 60.0
     config = ControllerConfig(
-        global_accelerator=GlobalAcceleratorConfig(workers=workers, cluster_name=cluster_name),
-        route53=Route53Config(workers=workers, cluster_name=cluster_name),
-        endpoint_group_binding=EndpointGroupBindingConfig(workers=workers),
+        global_accelerator=GlobalAcceleratorConfig(
+            workers=workers, cluster_name=cluster_name,
+            cloud_resync_period=cloud_resync,
+        ),
+        route53=Route53Config(
+            workers=workers, cluster_name=cluster_name,
+            cloud_resync_period=cloud_resync,
+        ),
+        endpoint_group_binding=EndpointGroupBindingConfig(
+            workers=workers, cloud_resync_period=cloud_resync,
+        ),
     )
     stop = setup_signal_handler()
 

@@ -101,6 +101,42 @@ def spawn_workers(
     return threads
 
 
+def spawn_cloud_resync(
+    period: float,
+    stop: threading.Event,
+    sources,
+    name: str,
+) -> threading.Thread | None:
+    """Opt-in periodic re-enqueue of ALL managed objects, even unchanged
+    ones (beyond-reference drift repair).
+
+    The reference's update handlers skip reflect.DeepEqual pairs
+    (ga/controller.go:99-101), so its 30 s informer resync never re-enqueues
+    an unchanged object — cloud-side drift on an object nobody edits is
+    never repaired (documented in docs/PARITY.md §resync).  With
+    ``period > 0`` this loop walks the informer caches every ``period``
+    seconds and re-enqueues every object passing the controller's
+    steady-state filter, so drifted cloud state converges back within one
+    period.  ``sources`` is a list of (list_fn, filter_fn, enqueue_fn).
+    Default 0 keeps exact reference parity (disabled)."""
+    if not period or period <= 0:
+        return None
+
+    def loop():
+        while not stop.wait(period):
+            for list_fn, filter_fn, enqueue_fn in sources:
+                try:
+                    for obj in list_fn():
+                        if filter_fn(obj):
+                            enqueue_fn(obj)
+                except Exception:
+                    logger.exception("cloud resync sweep for %s failed", name)
+
+    thread = threading.Thread(target=loop, name=f"{name}-cloud-resync", daemon=True)
+    thread.start()
+    return thread
+
+
 def make_queue_rate_limiter(qps: float, burst: int):
     """Controller queue limiter: per-item exponential backoff + overall
     token bucket (client-go DefaultControllerRateLimiter shape) with

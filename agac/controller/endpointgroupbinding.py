@@ -36,7 +36,7 @@ from ..kube.events import EventRecorder
 from ..kube.informer import wait_for_cache_sync
 from ..kube.store import is_not_found
 from ..kube.workqueue import RateLimitingQueue
-from .base import make_queue_rate_limiter, spawn_workers
+from .base import make_queue_rate_limiter, spawn_cloud_resync, spawn_workers
 
 logger = logging.getLogger(__name__)
 
@@ -48,6 +48,8 @@ class EndpointGroupBindingConfig:
     workers: int = 1
     queue_qps: float = 10.0
     queue_burst: int = 100
+    # opt-in drift repair (see docs/PARITY.md §resync); 0 = parity
+    cloud_resync_period: float = 0.0
 
 
 class EndpointGroupBindingController:
@@ -56,6 +58,7 @@ class EndpointGroupBindingController:
     delete_drain_requeue = 1.0
 
     def __init__(self, kube_client, informer_factory, config, cloud_factory):
+        self.cloud_resync_period = config.cloud_resync_period
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
@@ -97,6 +100,12 @@ class EndpointGroupBindingController:
                 return  # shutdown requested before caches synced
             raise RuntimeError("failed to wait for caches to sync")
         spawn_workers(threadiness, self._run_worker, CONTROLLER_AGENT_NAME, stop)
+        spawn_cloud_resync(
+            self.cloud_resync_period,
+            stop,
+            [(self.binding_lister.list, lambda o: True, self._enqueue)],
+            CONTROLLER_AGENT_NAME,
+        )
         stop.wait()
         self.workqueue.shut_down()
 

@@ -27,6 +27,7 @@ from ..kube.informer import wait_for_cache_sync
 from ..kube.workqueue import RateLimitingQueue
 from .base import (
     make_queue_rate_limiter,
+    spawn_cloud_resync,
     has_managed_annotation,
     managed_annotation_changed,
     objects_equal,
@@ -47,11 +48,16 @@ class GlobalAcceleratorConfig:
     # queue token-bucket rate (client-go default 10/100); raise for scale
     queue_qps: float = 10.0
     queue_burst: int = 100
+    # opt-in drift repair: re-enqueue unchanged managed objects every N
+    # seconds (0 = reference-parity behavior: cloud drift on an unchanged
+    # object is never repaired — see docs/PARITY.md §resync)
+    cloud_resync_period: float = 0.0
 
 
 class GlobalAcceleratorController:
     def __init__(self, kube_client, informer_factory, config, cloud_factory):
         self.cluster_name = config.cluster_name
+        self.cloud_resync_period = config.cloud_resync_period
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
         # (resource, ns, name) -> accelerator ARN hint.  Purely an API-cost
@@ -137,6 +143,23 @@ class GlobalAcceleratorController:
         logger.info("Starting workers")
         spawn_workers(threadiness, self._run_service_worker, CONTROLLER_AGENT_NAME + "-service", stop)
         spawn_workers(threadiness, self._run_ingress_worker, CONTROLLER_AGENT_NAME + "-ingress", stop)
+        spawn_cloud_resync(
+            self.cloud_resync_period,
+            stop,
+            [
+                (
+                    self.service_lister.list,
+                    lambda o: was_load_balancer_service(o) and has_managed_annotation(o),
+                    self._enqueue_service,
+                ),
+                (
+                    self.ingress_lister.list,
+                    lambda o: was_alb_ingress(o) and has_managed_annotation(o),
+                    self._enqueue_ingress,
+                ),
+            ],
+            CONTROLLER_AGENT_NAME,
+        )
         stop.wait()
         logger.info("Shutting down workers")
         self.service_queue.shut_down()

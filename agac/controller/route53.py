@@ -31,6 +31,7 @@ from .base import (
     has_hostname_annotation,
     hostname_annotation_changed,
     objects_equal,
+    spawn_cloud_resync,
     spawn_workers,
     was_load_balancer_service,
 )
@@ -47,11 +48,14 @@ class Route53Config:
     # queue token-bucket rate (client-go default 10/100); raise for scale
     queue_qps: float = 10.0
     queue_burst: int = 100
+    # opt-in drift repair (see docs/PARITY.md §resync); 0 = parity
+    cloud_resync_period: float = 0.0
 
 
 class Route53Controller:
     def __init__(self, kube_client, informer_factory, config, cloud_factory):
         self.cluster_name = config.cluster_name
+        self.cloud_resync_period = config.cloud_resync_period
         self.kube_client = kube_client
         self.cloud_factory = cloud_factory
         # LB hostname -> accelerator ARN hint (tag-verified before use;
@@ -143,6 +147,23 @@ class Route53Controller:
             raise RuntimeError("failed to wait for caches to sync")
         spawn_workers(threadiness, self._run_service_worker, CONTROLLER_AGENT_NAME + "-service", stop)
         spawn_workers(threadiness, self._run_ingress_worker, CONTROLLER_AGENT_NAME + "-ingress", stop)
+        spawn_cloud_resync(
+            self.cloud_resync_period,
+            stop,
+            [
+                (
+                    self.service_lister.list,
+                    lambda o: was_load_balancer_service(o) and has_hostname_annotation(o),
+                    self._enqueue_service,
+                ),
+                (
+                    self.ingress_lister.list,
+                    has_hostname_annotation,
+                    self._enqueue_ingress,
+                ),
+            ],
+            CONTROLLER_AGENT_NAME,
+        )
         stop.wait()
         self.service_queue.shut_down()
         self.ingress_queue.shut_down()
