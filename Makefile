@@ -36,3 +36,4 @@ run-webhook:
 
 manifests-validate:
 	$(PYTHON) -c "import yaml,glob; [list(yaml.safe_load_all(open(f))) for f in glob.glob('config/**/*.yaml', recursive=True)]; print('manifests OK')"
+	$(PYTHON) hack/helm_render.py --validate
