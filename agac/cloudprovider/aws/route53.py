@@ -13,6 +13,7 @@ import logging
 from typing import List, Optional, Tuple
 
 from ... import metrics
+from . import errors as awserr
 from . import types as t
 
 logger = logging.getLogger(__name__)
@@ -343,7 +344,22 @@ class Route53Mixin:
         )
 
     def _create_record_set(self, zone: t.HostedZone, hostname: str, accelerator):
-        self._change(zone, t.CHANGE_ACTION_CREATE, self._alias_record_set(hostname, accelerator))
+        try:
+            self._change(
+                zone, t.CHANGE_ACTION_CREATE, self._alias_record_set(hostname, accelerator)
+            )
+        except awserr.InvalidChangeBatch:
+            # Partial-create recovery (deliberate fix, docs/PARITY.md): a
+            # previous attempt may have committed the A record before a
+            # transient failure.  This path is only reached after TXT
+            # ownership of the name is established, so the A alias at the
+            # name is ours — converge it with UPSERT instead of looping on
+            # InvalidChangeBatch forever (the reference would loop:
+            # createRecordSet always uses CREATE, route53.go:240-264).
+            logger.info("A record for %s already exists; upserting", hostname)
+            self._change(
+                zone, t.CHANGE_ACTION_UPSERT, self._alias_record_set(hostname, accelerator)
+            )
 
     def _update_record_set(self, zone: t.HostedZone, hostname: str, accelerator):
         self._change(zone, t.CHANGE_ACTION_UPSERT, self._alias_record_set(hostname, accelerator))
@@ -351,16 +367,35 @@ class Route53Mixin:
     def _create_metadata_record_set(
         self, zone: t.HostedZone, hostname: str, owner_value: str
     ):
-        self._change(
-            zone,
-            t.CHANGE_ACTION_CREATE,
-            t.ResourceRecordSet(
-                name=hostname,
-                type=t.RR_TYPE_TXT,
-                ttl=TXT_TTL,
-                resource_records=[t.ResourceRecord(value=owner_value)],
-            ),
-        )
+        try:
+            self._change(
+                zone,
+                t.CHANGE_ACTION_CREATE,
+                t.ResourceRecordSet(
+                    name=hostname,
+                    type=t.RR_TYPE_TXT,
+                    ttl=TXT_TTL,
+                    resource_records=[t.ResourceRecord(value=owner_value)],
+                ),
+            )
+        except awserr.InvalidChangeBatch:
+            # Partial-create recovery: idempotent success ONLY if the
+            # existing TXT carries our ownership value; anything else means
+            # the name belongs to someone — re-raise (no stealing).
+            target = hostname if hostname.endswith(".") else hostname + "."
+            target = target.replace("*", "\\052", 1)
+            page, _ = self.route53.list_resource_record_sets(
+                zone.id, max_items=10, start_record_name=hostname
+            )
+            for rs in page:
+                if (
+                    rs.name == target
+                    and rs.type == t.RR_TYPE_TXT
+                    and any(r.value == owner_value for r in rs.resource_records)
+                ):
+                    logger.info("TXT ownership for %s already committed", hostname)
+                    return
+            raise
 
     def _delete_record(self, zone: t.HostedZone, record: t.ResourceRecordSet):
         self._change(zone, t.CHANGE_ACTION_DELETE, record)

@@ -49,6 +49,7 @@ def _paginate(items: list, max_results: Optional[int], token: Optional[str]):
 class FakeELBv2:
     def __init__(self, lock: threading.RLock):
         self._lock = lock
+        self.fault_hook = None  # see FakeGlobalAccelerator.fault_hook
         self._lbs: Dict[str, t.LoadBalancer] = {}  # arn -> LB
         self._regions: Dict[str, str] = {}  # arn -> region
         self._by_name: Dict[Tuple[str, str], str] = {}  # (region, name) -> arn
@@ -109,6 +110,8 @@ class FakeELBv2:
         marker: Optional[str] = None,
         page_size: Optional[int] = None,
     ) -> Tuple[List[t.LoadBalancer], Optional[str]]:
+        if self.fault_hook is not None:
+            self.fault_hook("elbv2", "describe_load_balancers")
         with self._lock:
             lbs = sorted(self._lbs.values(), key=lambda x: x.load_balancer_arn)
             if names:
@@ -134,6 +137,8 @@ class RegionalELBv2View:
         self.region = region
 
     def describe_load_balancers(self, names=None, marker=None, page_size=None):
+        if self._elbv2.fault_hook is not None:
+            self._elbv2.fault_hook("elbv2", "describe_load_balancers")
         with self._elbv2._lock:
             if names:
                 found, missing = [], []
@@ -165,6 +170,10 @@ class FakeGlobalAccelerator:
         self._lock = lock
         self.deploy_after_describes = deploy_after_describes
         self.call_counts: Dict[str, int] = {}
+        # fault injection (chaos tests): callable (service, op) that may
+        # raise an AWSAPIError to simulate throttles/5xx; fires at the top
+        # of every operation, before any state mutation
+        self.fault_hook = None
         self._accelerators: Dict[str, t.Accelerator] = {}
         self._pending: Dict[str, int] = {}  # arn -> remaining IN_PROGRESS describes
         self._tags: Dict[str, Dict[str, str]] = {}
@@ -177,6 +186,8 @@ class FakeGlobalAccelerator:
 
     def _count(self, op: str):
         self.call_counts[op] = self.call_counts.get(op, 0) + 1
+        if self.fault_hook is not None:
+            self.fault_hook("globalaccelerator", op)
 
     # -- lifecycle helper --------------------------------------------------
     def _mutated(self, arn: str):
@@ -487,6 +498,7 @@ def _normalize_record_name(name: str) -> str:
 class FakeRoute53:
     def __init__(self, lock: threading.RLock):
         self._lock = lock
+        self.fault_hook = None  # see FakeGlobalAccelerator.fault_hook
         self._zones: Dict[str, t.HostedZone] = {}
         # zone id -> {(name, type) -> ResourceRecordSet}
         self._records: Dict[str, Dict[Tuple[str, str], t.ResourceRecordSet]] = {}
@@ -503,6 +515,8 @@ class FakeRoute53:
 
     # -- API surface -------------------------------------------------------
     def list_hosted_zones(self, max_items: Optional[int] = None, marker=None):
+        if self.fault_hook is not None:
+            self.fault_hook("route53", "list_hosted_zones")
         with self._lock:
             zones = sorted(self._zones.values(), key=lambda z: z.name)
             page, token = _paginate(zones, max_items, marker)
@@ -510,6 +524,8 @@ class FakeRoute53:
 
     def list_hosted_zones_by_name(self, dns_name: str, max_items: Optional[int] = None):
         """Zones with name lexicographically >= dns_name, like the real API."""
+        if self.fault_hook is not None:
+            self.fault_hook("route53", "list_hosted_zones_by_name")
         with self._lock:
             zones = sorted(self._zones.values(), key=lambda z: z.name)
             after = [z for z in zones if z.name >= dns_name]
@@ -521,6 +537,8 @@ class FakeRoute53:
     ):
         """start_record_name mirrors the real API's StartRecordName: begin
         at the first record whose name >= the (normalized) given name."""
+        if self.fault_hook is not None:
+            self.fault_hook("route53", "list_resource_record_sets")
         with self._lock:
             records = self._records.get(zone_id)
             if records is None:
@@ -533,6 +551,8 @@ class FakeRoute53:
             return [_copy(r) for r in page], token
 
     def change_resource_record_sets(self, zone_id: str, changes: List[t.Change]):
+        if self.fault_hook is not None:
+            self.fault_hook("route53", "change_resource_record_sets")
         with self._lock:
             records = self._records.get(zone_id)
             if records is None:
@@ -574,6 +594,14 @@ class FakeAWSBackend:
         self.elbv2 = FakeELBv2(lock)
         self.ga = FakeGlobalAccelerator(lock, deploy_after_describes)
         self.route53 = FakeRoute53(lock)
+
+    def set_fault_hook(self, hook):
+        """Install ``hook(service, op)`` on every service; it may raise an
+        AWSAPIError to simulate throttling/5xx faults.  Fires before any
+        state mutation, so a failed call never half-applies.  None clears."""
+        self.elbv2.fault_hook = hook
+        self.ga.fault_hook = hook
+        self.route53.fault_hook = hook
 
 
 def _copy(obj):
