@@ -45,8 +45,6 @@ def apply_document(client, doc: dict) -> Tuple[str, str]:
     except NotFoundError:
         client.create(obj)
         return "created", ident
-    # adopt the live resourceVersion, keep the manifest's spec/metadata
-    obj.metadata.resource_version = existing.metadata.resource_version
     from ..apis.meta import to_dict
 
     desired, live = to_dict(obj), to_dict(existing)
@@ -56,7 +54,24 @@ def apply_document(client, doc: dict) -> Tuple[str, str]:
     )
     if unchanged:
         return "unchanged", ident
-    client.update(obj)
+    # PATCH the manifest's fields instead of full PUT replacement, so
+    # server-populated fields (status, uid, creationTimestamp, fields other
+    # managers own) survive — the merge-patch analogue of the reference
+    # e2e's server-side apply (e2e/pkg/util/manifests.go:35-70); the k8s
+    # wire client stamps ?fieldManager= on the request.
+    patch = {
+        "metadata": {
+            "labels": desired.get("metadata", {}).get("labels"),
+            "annotations": desired.get("metadata", {}).get("annotations"),
+        },
+        "spec": desired.get("spec"),
+    }
+    patch["metadata"] = {k: v for k, v in patch["metadata"].items() if v is not None}
+    if not patch["metadata"]:
+        del patch["metadata"]
+    if patch.get("spec") is None:
+        patch.pop("spec", None)
+    client.patch(kind, namespace, obj.metadata.name, patch)
     return "configured", ident
 
 

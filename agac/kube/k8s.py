@@ -114,12 +114,17 @@ class K8sKubeClient(KubeClient):
     shedding) up to ``max_retries`` times."""
 
     def __init__(self, config: RestConfig, timeout: float = 10.0,
-                 page_size: int = 500, max_retries: int = 5):
+                 page_size: int = 500, max_retries: int = 5,
+                 field_manager: str = "aws-global-accelerator-controller"):
         self.config = config
         self.base_url = config.host
         self.timeout = timeout
         self.page_size = page_size
         self.max_retries = max_retries
+        # declared owner of fields we write (apiserver server-side field
+        # tracking; sent as ?fieldManager= on every mutating verb, the
+        # client-go behavior the reference inherits)
+        self.field_manager = field_manager
         self.session = requests.Session()
         if config.token:
             self.session.headers["Authorization"] = f"Bearer {config.token}"
@@ -170,6 +175,7 @@ class K8sKubeClient(KubeClient):
         r = self._request(
             "post",
             self._url(kind, obj.metadata.namespace or None),
+            params={"fieldManager": self.field_manager},
             json=self._obj_body(obj),
             timeout=self.timeout,
         )
@@ -227,6 +233,7 @@ class K8sKubeClient(KubeClient):
         r = self._request(
             "put",
             self._url(kind, obj.metadata.namespace or None, obj.metadata.name),
+            params={"fieldManager": self.field_manager},
             json=self._obj_body(obj),
             timeout=self.timeout,
         )
@@ -238,6 +245,7 @@ class K8sKubeClient(KubeClient):
         r = self._request(
             "put",
             self._url(kind, obj.metadata.namespace or None, obj.metadata.name, "status"),
+            params={"fieldManager": self.field_manager},
             json=self._obj_body(obj),
             timeout=self.timeout,
         )
@@ -258,6 +266,7 @@ class K8sKubeClient(KubeClient):
         r = self._request(
             "patch",
             self._url(kind, namespace or None, name, subresource),
+            params={"fieldManager": self.field_manager},
             json=patch,
             headers={"Content-Type": MERGE_PATCH_CONTENT_TYPE},
             timeout=self.timeout,
