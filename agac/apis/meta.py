@@ -128,36 +128,125 @@ class TypeMeta:
 
 
 _SCALARS = (str, int, float, bool, bytes, type(None))
+_SCALAR_TYPES = {str, int, float, bool, bytes, type(None)}
 
-# per-class field-name cache for the fast copier
-_FIELDS_CACHE: dict = {}
+# per-class compiled copier cache
+_COPIER_CACHE: dict = {}
 
 
-def deep_copy(obj):
-    """DeepCopy equivalent (reference uses generated DeepCopyObject).
+def _classify_hint(hint):
+    """Map a field's type annotation to a copy strategy tag.
+    Returns (tag, subclass|None); tag ∈ scalar / scalar_dict / scalar_list /
+    dataclass / dataclass_list / fallback."""
+    hint = _unwrap_optional(hint)
+    if hint in _SCALAR_TYPES:
+        return "scalar", None
+    origin = typing.get_origin(hint)
+    if origin is dict:
+        args = typing.get_args(hint)
+        if len(args) == 2 and args[1] in _SCALAR_TYPES:
+            return "scalar_dict", None
+        return "fallback", None
+    if origin is list:
+        args = typing.get_args(hint)
+        if len(args) == 1:
+            if args[0] in _SCALAR_TYPES:
+                return "scalar_list", None
+            if _is_api_type(args[0]):
+                return "dataclass_list", args[0]
+        return "fallback", None
+    if _is_api_type(hint):
+        return "dataclass", hint
+    return "fallback", None
 
-    Hand-rolled recursion instead of copy.deepcopy: API objects are plain
-    dataclass trees of scalars/lists/dicts, and this path is the hottest
-    allocation site in the whole framework (every store read/write and
-    informer dispatch copies).  ~6x faster than copy.deepcopy here.
-    """
+
+def _compile_copier(cls):
+    """exec-generate a specialized copier for one dataclass: scalar fields
+    assign directly, scalar dicts/lists shallow-copy (values immutable),
+    nested API types recurse through their own compiled copier.  This is
+    the framework's hottest path (the client-go generated-DeepCopy
+    analogue): every store read/write and informer dispatch copies."""
+    try:
+        hints = typing.get_type_hints(cls)
+    except Exception:
+        hints = {}
+    env = {"_cls": cls, "_new": object.__new__, "_deep": deep_copy}
+    lines = ["def _copy(o):", "    n = _new(_cls)"]
+    for f in dataclasses.fields(cls):
+        name = f.name
+        tag, sub = _classify_hint(hints.get(f.name, object))
+        if tag == "scalar":
+            lines.append(f"    n.{name} = o.{name}")
+        elif tag == "scalar_dict":
+            lines.append(f"    v = o.{name}; n.{name} = dict(v) if v is not None else None")
+        elif tag == "scalar_list":
+            lines.append(f"    v = o.{name}; n.{name} = list(v) if v is not None else None")
+        elif tag == "dataclass":
+            sub_copier = _copier_for(sub)
+            env[f"_c_{name}"] = sub_copier
+            lines.append(
+                f"    v = o.{name}; n.{name} = _c_{name}(v) if v is not None else None"
+            )
+        elif tag == "dataclass_list":
+            sub_copier = _copier_for(sub)
+            env[f"_c_{name}"] = sub_copier
+            lines.append(
+                f"    v = o.{name}; "
+                f"n.{name} = [_c_{name}(i) for i in v] if v is not None else None"
+            )
+        else:
+            lines.append(f"    n.{name} = _deep(o.{name})")
+    lines.append("    return n")
+    exec("\n".join(lines), env)  # noqa: S102 - trusted, generated from dataclass fields
+    return env["_copy"]
+
+
+def _copier_for(cls):
+    copier = _COPIER_CACHE.get(cls)
+    if copier is None:
+        # placeholder guards against recursive class graphs (none today)
+        _COPIER_CACHE[cls] = lambda o: _generic_copy(o)
+        copier = _compile_copier(cls)
+        _COPIER_CACHE[cls] = copier
+    return copier
+
+
+def _generic_copy(obj):
     if isinstance(obj, _SCALARS):
         return obj
     if isinstance(obj, list):
         return [deep_copy(v) for v in obj]
     if isinstance(obj, dict):
         return {k: deep_copy(v) for k, v in obj.items()}
+    if dataclasses.is_dataclass(obj):
+        new = obj.__class__.__new__(obj.__class__)
+        for f in dataclasses.fields(obj):
+            setattr(new, f.name, deep_copy(getattr(obj, f.name)))
+        return new
+    return copy.deepcopy(obj)
+
+
+def deep_copy(obj):
+    """DeepCopy equivalent (reference uses generated DeepCopyObject).
+
+    Per-class copiers are exec-compiled on first use from the dataclass's
+    type annotations (scalars assigned, scalar containers shallow-copied,
+    nested API types recursed) — the same codegen idea as the reference's
+    zz_generated.deepcopy.go, done at runtime.  Fields holding values that
+    don't match their annotation fall back to a generic recursive copy."""
     cls = obj.__class__
-    names = _FIELDS_CACHE.get(cls)
-    if names is None:
-        if not dataclasses.is_dataclass(obj):
-            return copy.deepcopy(obj)
-        names = [f.name for f in dataclasses.fields(cls)]
-        _FIELDS_CACHE[cls] = names
-    new = cls.__new__(cls)
-    for name in names:
-        setattr(new, name, deep_copy(getattr(obj, name)))
-    return new
+    copier = _COPIER_CACHE.get(cls)
+    if copier is not None:
+        return copier(obj)
+    if isinstance(obj, _SCALARS):
+        return obj
+    if cls is list:
+        return [deep_copy(v) for v in obj]
+    if cls is dict:
+        return {k: deep_copy(v) for k, v in obj.items()}
+    if dataclasses.is_dataclass(obj) and not isinstance(obj, type):
+        return _copier_for(cls)(obj)
+    return copy.deepcopy(obj)
 
 
 def meta_namespace_key(obj) -> str:
