@@ -174,18 +174,35 @@ def run_step(client, backend, services, step_idx: int, timeout: float = 300.0,
              bindings=()):
     """Mutate every service's port (and every binding's weight in the full
     scenario) and wait for full convergence."""
+    from agac.kube.store import ConflictError
+
+    def update_with_retry(kind, name, mutate, attempts=10):
+        """RetryOnConflict (client-go util/retry): the controllers update
+        status concurrently, so a spec write can race a status write."""
+        for _ in range(attempts):
+            obj = client.get(kind, "default", name)
+            mutate(obj)
+            try:
+                client.update(obj)
+                return
+            except ConflictError:
+                continue
+        raise RuntimeError(f"{kind}/{name}: conflict retry budget exhausted")
+
     port = 8000 + (step_idx % 2)
     weight = 100 + (step_idx % 2)
     owner_to_port = {}
     for name in services:
-        svc = client.get("Service", "default", name)
-        svc.spec.ports[0].port = port
-        client.update(svc)
+        def set_port(svc):
+            svc.spec.ports[0].port = port
+
+        update_with_retry("Service", name, set_port)
         owner_to_port[f"service/default/{name}"] = port
     for name in bindings:
-        binding = client.get("EndpointGroupBinding", "default", name)
-        binding.spec.weight = weight
-        client.update(binding)
+        def set_weight(binding):
+            binding.spec.weight = weight
+
+        update_with_retry("EndpointGroupBinding", name, set_weight)
     deadline = time.monotonic() + timeout
     # adaptive poll: the convergence check is O(objects) under the backend
     # lock, so polling every 1ms would contend with the workers at scale
