@@ -20,6 +20,10 @@ logger = logging.getLogger(__name__)
 
 DEFAULT_GA_MISSING_RETRY = 60.0  # seconds (reference route53.go:72-76)
 TXT_TTL = 300  # seconds (reference :276)
+# hostname->zone hint lifetime: after this the parent-domain walk re-runs,
+# so zone-topology changes (new more-specific zone) are honored within one
+# TTL instead of only on NoSuchHostedZone (same magnitude as the TXT TTL)
+ZONE_HINT_TTL = 300.0
 
 
 # ---------------------------------------------------------------------------
@@ -77,10 +81,16 @@ class Route53Mixin:
     def ensure_route53_for_service(
         self, svc, lb_ingress, hostnames: List[str], cluster_name: str,
         hint_arn: Optional[str] = None,
+        zone_hints: Optional[dict] = None,
     ) -> Tuple[bool, float]:
         """``hint_arn`` short-circuits the by-hostname accelerator scan with
         tag-verified fallback (same contract as
-        ensure_global_accelerator_for_service's hint)."""
+        ensure_global_accelerator_for_service's hint).  ``zone_hints`` is a
+        caller-owned hostname→HostedZone cache that skips the parent-domain
+        ListHostedZonesByName walk; a hinted zone that turns out to be gone
+        (NoSuchHostedZone on any operation) is dropped and the walk re-runs,
+        so behavior is identical to the reference's per-reconcile walk
+        (route53.go:335-358)."""
         return self._ensure_route53(
             lb_ingress.hostname,
             hostnames,
@@ -89,11 +99,13 @@ class Route53Mixin:
             svc.metadata.namespace,
             svc.metadata.name,
             hint_arn=hint_arn,
+            zone_hints=zone_hints,
         )
 
     def ensure_route53_for_ingress(
         self, ingress, lb_ingress, hostnames: List[str], cluster_name: str,
         hint_arn: Optional[str] = None,
+        zone_hints: Optional[dict] = None,
     ) -> Tuple[bool, float]:
         return self._ensure_route53(
             lb_ingress.hostname,
@@ -103,6 +115,7 @@ class Route53Mixin:
             ingress.metadata.namespace,
             ingress.metadata.name,
             hint_arn=hint_arn,
+            zone_hints=zone_hints,
         )
 
     def _verified_hostname_hint(self, hint_arn: str, lb_hostname: str, cluster_name: str):
@@ -141,6 +154,7 @@ class Route53Mixin:
         ns: str,
         name: str,
         hint_arn: Optional[str] = None,
+        zone_hints: Optional[dict] = None,
     ) -> Tuple[bool, float]:
         """Returns (created, retry_after_seconds).  0 or >1 matching
         accelerators ⇒ requeue after 60s (reference :62-78)."""
@@ -166,30 +180,66 @@ class Route53Mixin:
 
         owner_value = route53_owner_value(cluster_name, resource, ns, name)
         created = False
+        import time as _time
+
         for hostname in hostnames:
-            hosted_zone = self.get_hosted_zone(hostname)
+            hinted = None
+            if zone_hints is not None:
+                entry = zone_hints.get(hostname)
+                if entry is not None:
+                    zone, stamp = entry
+                    # TTL bounds divergence from the reference's
+                    # per-reconcile walk: a newly-created more-specific
+                    # hosted zone is picked up within ZONE_HINT_TTL
+                    if _time.monotonic() - stamp < ZONE_HINT_TTL:
+                        hinted = zone
+                    else:
+                        zone_hints.pop(hostname, None)
+            hosted_zone = hinted or self.get_hosted_zone(hostname)
             logger.info("HostedZone is %s", hosted_zone.id)
-            record = self._find_owned_a_record_at(hosted_zone, hostname, owner_value)
-            if record is None:
-                logger.info(
-                    "Creating record for %s with %s",
-                    hostname,
-                    accelerator.accelerator_arn,
+            try:
+                did_create = self._sync_one_hostname(
+                    hosted_zone, hostname, owner_value, accelerator
                 )
-                self._create_metadata_record_set(
-                    hosted_zone, hostname, owner_value
+            except awserr.NoSuchHostedZone:
+                if hinted is None:
+                    raise
+                # stale zone hint (zone deleted/recreated): drop it and
+                # redo the reference's parent-domain walk once
+                zone_hints.pop(hostname, None)
+                hinted = None
+                hosted_zone = self.get_hosted_zone(hostname)
+                did_create = self._sync_one_hostname(
+                    hosted_zone, hostname, owner_value, accelerator
                 )
-                self._create_record_set(hosted_zone, hostname, accelerator)
-                created = True
-            else:
-                if not need_records_update(record, accelerator):
-                    logger.info("Do not need to update for %s, so skip it", record.name)
-                    continue
-                self._update_record_set(hosted_zone, hostname, accelerator)
-                logger.info("RecordSet %s is updated", record.name)
+            created = created or did_create
+            if zone_hints is not None and hinted is None:
+                zone_hints[hostname] = (hosted_zone, _time.monotonic())
 
         logger.info("All records are synced for %s %s/%s", resource, ns, name)
         return created, 0.0
+
+    def _sync_one_hostname(
+        self, hosted_zone: t.HostedZone, hostname: str, owner_value: str, accelerator
+    ) -> bool:
+        """Create or drift-repair the TXT+A pair for one hostname in one
+        zone; returns True if records were created."""
+        record = self._find_owned_a_record_at(hosted_zone, hostname, owner_value)
+        if record is None:
+            logger.info(
+                "Creating record for %s with %s",
+                hostname,
+                accelerator.accelerator_arn,
+            )
+            self._create_metadata_record_set(hosted_zone, hostname, owner_value)
+            self._create_record_set(hosted_zone, hostname, accelerator)
+            return True
+        if not need_records_update(record, accelerator):
+            logger.info("Do not need to update for %s, so skip it", record.name)
+            return False
+        self._update_record_set(hosted_zone, hostname, accelerator)
+        logger.info("RecordSet %s is updated", record.name)
+        return False
 
     def cleanup_record_set(
         self, cluster_name: str, resource: str, ns: str, name: str

@@ -61,6 +61,10 @@ class Route53Controller:
         # LB hostname -> accelerator ARN hint (tag-verified before use;
         # misses fall back to the full by-hostname scan)
         self._arn_hints = {}
+        # hostname -> HostedZone hint: skips the parent-domain zone walk;
+        # a stale entry (zone gone) is dropped by the resource manager and
+        # the walk re-runs (see Route53Mixin.ensure_route53_for_service)
+        self._zone_hints = {}
         self._hints_lock = threading.Lock()
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
         self.service_queue = RateLimitingQueue(
@@ -236,6 +240,7 @@ class Route53Controller:
             created, retry_after = cloud.ensure_route53_for_service(
                 svc, lb_ingress, hostnames, self.cluster_name,
                 hint_arn=self._hint_for(lb_ingress.hostname),
+                zone_hints=self._zone_hints,
             )
             if retry_after == 0:
                 self._remember_hint(lb_ingress.hostname, cloud)
@@ -300,6 +305,7 @@ class Route53Controller:
             created, retry_after = cloud.ensure_route53_for_ingress(
                 ingress, lb_ingress, hostnames, self.cluster_name,
                 hint_arn=self._hint_for(lb_ingress.hostname),
+                zone_hints=self._zone_hints,
             )
             if retry_after == 0:
                 self._remember_hint(lb_ingress.hostname, cloud)

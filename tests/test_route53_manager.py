@@ -274,3 +274,81 @@ class TestOwnedRecordDiscoveryTyping:
         aaaa = recs[("www.example.com.", "AAAA")]
         assert aaaa.alias_target.dns_name.rstrip(".") == "stale.awsglobalaccelerator.com"
         assert ("www.example.com.", "A") in recs
+
+
+class TestZoneHints:
+    """Controller-held hostname→zone cache skips the parent-domain walk;
+    stale hints self-heal via NoSuchHostedZone fallback and a TTL."""
+
+    def _counting_hook(self, backend):
+        counts = {}
+
+        def hook(service, op):
+            counts[(service, op)] = counts.get((service, op), 0) + 1
+
+        backend.set_fault_hook(hook)
+        return counts
+
+    def test_hint_skips_zone_walk(self, backend, cloud):
+        backend.route53.create_hosted_zone("example.com")
+        lb = backend.elbv2.create_load_balancer("mylb", region=REGION)
+        seed_ga_for_lb(backend, cloud, lb)
+        hints = {}
+        cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.example.com"], CLUSTER, zone_hints=hints,
+        )
+        assert "www.example.com" in hints
+        counts = self._counting_hook(backend)
+        cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.example.com"], CLUSTER, zone_hints=hints,
+        )
+        assert counts.get(("route53", "list_hosted_zones_by_name"), 0) == 0
+
+    def test_stale_hint_falls_back_to_walk(self, backend, cloud):
+        zone = backend.route53.create_hosted_zone("example.com")
+        lb = backend.elbv2.create_load_balancer("mylb", region=REGION)
+        seed_ga_for_lb(backend, cloud, lb)
+        hints = {}
+        cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.example.com"], CLUSTER, zone_hints=hints,
+        )
+        # zone deleted and recreated under a new id
+        del backend.route53._zones[zone.id]
+        del backend.route53._records[zone.id]
+        new_zone = backend.route53.create_hosted_zone("example.com")
+        created, retry = cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.example.com"], CLUSTER, zone_hints=hints,
+        )
+        assert created and retry == 0
+        recs, _ = backend.route53.list_resource_record_sets(new_zone.id)
+        assert {("www.example.com.", "A"), ("www.example.com.", "TXT")} <= {
+            (r.name, r.type) for r in recs
+        }
+        assert hints["www.example.com"][0].id == new_zone.id
+
+    def test_hint_ttl_reruns_walk(self, backend, cloud, monkeypatch):
+        from agac.cloudprovider.aws import route53 as r53mod
+
+        backend.route53.create_hosted_zone("example.com")
+        lb = backend.elbv2.create_load_balancer("mylb", region=REGION)
+        seed_ga_for_lb(backend, cloud, lb)
+        hints = {}
+        cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.sub.example.com"], CLUSTER, zone_hints=hints,
+        )
+        # a more specific zone appears; the reference's per-reconcile walk
+        # would pick it — the hint must honor it within one TTL
+        specific = backend.route53.create_hosted_zone("sub.example.com")
+        monkeypatch.setattr(r53mod, "ZONE_HINT_TTL", 0.0)
+        cloud.ensure_route53_for_service(
+            mk_service(), corev1.LoadBalancerIngress(hostname=lb.dns_name),
+            ["www.sub.example.com"], CLUSTER, zone_hints=hints,
+        )
+        assert hints["www.sub.example.com"][0].id == specific.id
+        recs, _ = backend.route53.list_resource_record_sets(specific.id)
+        assert ("www.sub.example.com.", "A") in {(r.name, r.type) for r in recs}
