@@ -21,12 +21,24 @@ logger = logging.getLogger(__name__)
 
 
 class EventRecorder:
+    """Async, like client-go: record.EventBroadcaster buffers events on a
+    channel and a background goroutine writes them to the API (with
+    count-aggregation), so recording never blocks a reconcile on API I/O.
+    ``flush()`` drains the queue — call it in tests that assert on Event
+    objects immediately after recording."""
+
     def __init__(self, client, component: str):
+        import queue as _queue
+
         self._client = client
         self.component = component
-        self._lock = threading.Lock()
         # (kind, ns, name, type, reason, message) -> Event name
         self._seen = {}
+        self._queue: "_queue.Queue" = _queue.Queue(maxsize=1024)
+        self._thread = threading.Thread(
+            target=self._writer, name=f"event-recorder-{component}", daemon=True
+        )
+        self._thread.start()
 
     def event(self, obj, event_type: str, reason: str, message: str):
         ref = corev1.ObjectReference(
@@ -40,33 +52,56 @@ class EventRecorder:
             ref.namespace, ref.name, event_type, reason, message,
         )
         now = time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
-        key = (ref.kind, ref.namespace, ref.name, event_type, reason, message)
-        with self._lock:
-            existing_name = self._seen.get(key)
+        try:
+            self._queue.put_nowait((ref, event_type, reason, message, now))
+        except Exception:
+            # full buffer: drop, like client-go's broadcaster under pressure
+            logger.warning("event buffer full; dropping event %r", reason)
+
+    def _writer(self):
+        while True:
+            item = self._queue.get()
             try:
-                if existing_name is not None:
-                    ev = self._client.get("Event", ref.namespace, existing_name)
-                    ev.count += 1
-                    ev.last_timestamp = now
-                    self._client.update(ev)
-                    return
-                name = f"{ref.name}.{uuid.uuid4().hex[:10]}"
-                ev = corev1.Event(
-                    metadata=ObjectMeta(name=name, namespace=ref.namespace or "default"),
-                    involved_object=ref,
-                    reason=reason,
-                    message=message,
-                    type=event_type,
-                    source=corev1.EventSource(component=self.component),
-                    count=1,
-                    first_timestamp=now,
-                    last_timestamp=now,
-                )
-                self._client.create(ev)
-                self._seen[key] = name
+                self._write(*item)
             except Exception:
-                # Event recording must never break reconciliation.
-                logger.exception("failed to record event %r", reason)
+                # Event recording must never break anything.
+                logger.exception("failed to record event %r", item[2])
+            finally:
+                self._queue.task_done()
+
+    def _write(self, ref, event_type: str, reason: str, message: str, now: str):
+        key = (ref.kind, ref.namespace, ref.name, event_type, reason, message)
+        existing_name = self._seen.get(key)
+        if existing_name is not None:
+            ev = self._client.get("Event", ref.namespace, existing_name)
+            ev.count += 1
+            ev.last_timestamp = now
+            self._client.update(ev)
+            return
+        name = f"{ref.name}.{uuid.uuid4().hex[:10]}"
+        ev = corev1.Event(
+            metadata=ObjectMeta(name=name, namespace=ref.namespace or "default"),
+            involved_object=ref,
+            reason=reason,
+            message=message,
+            type=event_type,
+            source=corev1.EventSource(component=self.component),
+            count=1,
+            first_timestamp=now,
+            last_timestamp=now,
+        )
+        self._client.create(ev)
+        self._seen[key] = name
+
+    def flush(self, timeout: float = 5.0) -> bool:
+        """Block until every queued event has been written (or timeout).
+        Returns True if fully drained."""
+        deadline = time.monotonic() + timeout
+        while time.monotonic() < deadline:
+            if self._queue.unfinished_tasks == 0:
+                return True
+            time.sleep(0.005)
+        return self._queue.unfinished_tasks == 0
 
     def eventf(self, obj, event_type: str, reason: str, fmt: str, *args):
         self.event(obj, event_type, reason, fmt % args if args else fmt)

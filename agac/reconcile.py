@@ -24,7 +24,6 @@ import time
 from dataclasses import dataclass
 
 from . import metrics
-from .apis.meta import deep_copy
 from .errors import is_no_retry
 from .kube.store import is_not_found
 
@@ -73,7 +72,11 @@ def _reconcile_handler(key, queue, key_to_obj, process_delete, process_create_or
                 outcome = "error"
                 return
         else:
-            res, err = _run(process_create_or_update, deep_copy(obj))
+            # The reference DeepCopies here (keyToService returns the shared
+            # informer-cache pointer, reconcile.go:52).  Our listers already
+            # return a private deep copy per Get (informer.cache_get), so a
+            # second copy would be pure overhead — the process func owns obj.
+            res, err = _run(process_create_or_update, obj)
 
         if err is not None:
             if is_no_retry(err):

@@ -19,6 +19,7 @@ def test_event_creates_object_with_reference():
     client = InMemoryKubeClient()
     recorder = EventRecorder(client, "test-controller")
     recorder.event(mk_obj(), "Normal", "GlobalAcceleratorCreated", "created arn:x")
+    assert recorder.flush()
     events = list_events(client)
     assert len(events) == 1
     ev = events[0]
@@ -35,6 +36,7 @@ def test_repeated_event_aggregates_count():
     recorder = EventRecorder(client, "c")
     for _ in range(3):
         recorder.event(mk_obj(), "Normal", "Reason", "same message")
+    assert recorder.flush()
     events = list_events(client)
     assert len(events) == 1
     assert events[0].count == 3
@@ -45,6 +47,7 @@ def test_different_messages_make_distinct_events():
     recorder = EventRecorder(client, "c")
     recorder.eventf(mk_obj(), "Normal", "Reason", "msg %d", 1)
     recorder.eventf(mk_obj(), "Normal", "Reason", "msg %d", 2)
+    assert recorder.flush()
     assert len(list_events(client)) == 2
 
 
@@ -55,6 +58,30 @@ def test_recorder_never_raises():
 
     recorder = EventRecorder(BrokenClient(), "c")
     recorder.event(mk_obj(), "Normal", "Reason", "m")  # must not raise
+    assert recorder.flush()  # writer thread survives the failure
+
+
+def test_recording_is_async_and_flushable():
+    """client-go EventBroadcaster behavior: event() never blocks on API
+    I/O; flush() drains the buffer."""
+    import threading as _threading
+    import time as _time
+
+    gate = _threading.Event()
+
+    class SlowClient(InMemoryKubeClient):
+        def create(self, obj):
+            gate.wait(2.0)
+            return super().create(obj)
+
+    client = SlowClient()
+    recorder = EventRecorder(client, "c")
+    t0 = _time.monotonic()
+    recorder.event(mk_obj(), "Normal", "Reason", "m")
+    assert _time.monotonic() - t0 < 0.5  # did not block on the slow create
+    gate.set()
+    assert recorder.flush()
+    assert len(list_events(client)) == 1
 
 
 def test_metrics_counters_observable():

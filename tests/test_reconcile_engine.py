@@ -29,13 +29,45 @@ def pump(q, key_to_obj, on_delete, on_update):
     return process_next_work_item(q, key_to_obj, on_delete, on_update)
 
 
-def test_create_or_update_called_with_copy():
+def test_create_or_update_owns_the_looked_up_object():
+    """The copy-before-process contract moved to key_to_obj: listers return
+    a private deep copy per Get (informer.cache_get), so the engine passes
+    the lookup result through unchanged instead of copying a second time
+    (the reference copies in reconcile.go:52 because its lister returns the
+    shared cache pointer)."""
     q = make_queue()
     calls = []
     original = Obj("x")
     pump(q, lambda k: original, lambda k: Result(), lambda o: calls.append(o) or Result())
     assert len(calls) == 1
-    assert calls[0] is not original  # DeepCopyObject before processing
+    assert calls[0] is original  # no redundant second copy
+
+    # ...and the real key_to_obj path (a Lister) DOES hand out private
+    # copies, so mutations by a process func never leak into the cache
+    from agac.apis import core as corev1
+    from agac.apis.meta import ObjectMeta
+    from agac.kube.client import InMemoryKubeClient
+    from agac.kube.informer import Informer
+
+    client = InMemoryKubeClient()
+    client.create(corev1.Service(metadata=ObjectMeta(name="x", namespace="default")))
+    informer = Informer(client, "Service", resync_period=0)
+    import threading
+
+    stop = threading.Event()
+    informer.run(stop)
+    try:
+        from agac.kube.informer import wait_for_cache_sync
+
+        assert wait_for_cache_sync(stop, informer)
+        a = informer.lister().get("x", namespace="default")
+        b = informer.lister().get("x", namespace="default")
+        assert a is not b
+        a.metadata.annotations["mutated"] = "true"
+        assert "mutated" not in informer.lister().get("x", namespace="default").metadata.annotations
+    finally:
+        stop.set()
+        informer.stop()
 
 
 def test_not_found_routes_to_delete():
