@@ -66,7 +66,12 @@ class Informer:
 
     # -- registration ------------------------------------------------------
     def add_event_handler(self, on_add=None, on_update=None, on_delete=None):
-        """Handlers: on_add(obj), on_update(old, new), on_delete(obj)."""
+        """Handlers: on_add(obj), on_update(old, new), on_delete(obj).
+
+        READ-ONLY CONTRACT (same as client-go shared informers): handlers
+        receive the informer's cached objects directly — they MUST NOT
+        mutate them.  Controllers that need a mutable object go through a
+        Lister Get, which returns a private deep copy."""
         self._handlers.append((on_add, on_update, on_delete))
 
     def lister(self) -> Lister:
@@ -149,16 +154,18 @@ class Informer:
             fresh = {(o.metadata.namespace, o.metadata.name): o for o in items}
             old_cache = self._cache
             self._cache = fresh
-        # Deliver deltas vs the previous cache contents (first run: all adds).
+        # Deliver deltas vs the previous cache contents (first run: all
+        # adds).  Cached objects are handed to handlers directly — see the
+        # read-only contract on add_event_handler.
         for key, obj in fresh.items():
             old = old_cache.get(key)
             if old is None:
-                self._dispatch_add(metalib.deep_copy(obj))
+                self._dispatch_add(obj)
             else:
-                self._dispatch_update(metalib.deep_copy(old), metalib.deep_copy(obj))
+                self._dispatch_update(old, obj)
         for key, obj in old_cache.items():
             if key not in fresh:
-                self._dispatch_delete(metalib.deep_copy(obj))
+                self._dispatch_delete(obj)
         self._synced.set()
 
         # Watch loop: a closed stream re-watches from the last delivered
@@ -199,16 +206,23 @@ class Informer:
             with self._cache_lock:
                 old = self._cache.get(key)
                 self._cache[key] = obj
+            # handlers get the cached object itself (read-only contract)
             if old is None:
-                self._dispatch_add(metalib.deep_copy(obj))
+                self._dispatch_add(obj)
             else:
-                self._dispatch_update(old, metalib.deep_copy(obj))
+                self._dispatch_update(old, obj)
 
     def _resync_loop(self):
         stop = self._stop
         while not stop.wait(self.resync_period):
-            for obj in self.cache_list():
-                self._dispatch_update(metalib.deep_copy(obj), obj)
+            with self._cache_lock:
+                snapshot = list(self._cache.values())
+            for obj in snapshot:
+                # update(obj, obj) with the SAME cached object on both
+                # sides, exactly like client-go resync — the DeepEqual /
+                # rv-shortcut guard in the handlers drops these unless a
+                # controller opts into cloud-resync re-enqueue
+                self._dispatch_update(obj, obj)
 
 
 class SharedInformerFactory:
