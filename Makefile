@@ -11,9 +11,14 @@ test-fast:
 
 # threading stress profile: repeat the concurrency-heavy suites
 test-race:
+	# lock-order checking (agac/lockcheck.py): the concurrency suites run
+	# with every project lock instrumented; a lock-order cycle (potential
+	# deadlock) fails the session — plus 3x stress repetition for flake
+	# surfacing (the go-test-race analogue this pure-Python tier can have)
 	for i in 1 2 3; do \
-		$(PYTHON) -m pytest tests/test_stress_concurrency.py tests/test_chaos_recovery.py \
-			tests/test_store_event_sourcing.py -q || exit 1; \
+		AGAC_LOCKCHECK=1 $(PYTHON) -m pytest tests/test_stress_concurrency.py \
+			tests/test_chaos_recovery.py tests/test_store_event_sourcing.py \
+			tests/test_fault_injection.py tests/test_cloud_resync.py -q || exit 1; \
 	done
 
 bench:
