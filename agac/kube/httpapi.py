@@ -309,6 +309,12 @@ class _Handler(BaseHTTPRequestHandler):
                 bookmark_rv = self.store.latest_rv()
                 event = watch.get(timeout=self.watch_idle_seconds)
                 if event is None:
+                    if watch.closed:
+                        # the store-side subscription ended — end the HTTP
+                        # stream too so the client reconnects and resumes
+                        # from its last delivered rv (real apiservers close
+                        # watches periodically; clients must handle it)
+                        break
                     if bookmarks:
                         payload = {
                             "type": "BOOKMARK",
@@ -331,6 +337,11 @@ class _Handler(BaseHTTPRequestHandler):
                 if not k8s_style:
                     payload["resourceVersion"] = event.resource_version
                 self._write_chunk(json.dumps(payload).encode() + b"\n")
+            # clean end-of-stream: terminate the chunked body so the client
+            # sees EOF promptly instead of waiting for a socket teardown
+            self.wfile.write(b"0\r\n\r\n")
+            self.wfile.flush()
+            self.close_connection = True
         except (BrokenPipeError, ConnectionResetError, OSError):
             pass
         finally:
