@@ -180,3 +180,85 @@ def test_ingress_wire_roundtrip_stable(ingress):
     d1 = to_dict(ingress)
     back = from_dict(corev1.Ingress, d1)
     assert to_dict(back) == d1
+
+
+# ---------------------------------------------------------------------------
+# Paginated-list invariants (round 2: limit/continue chunking)
+# ---------------------------------------------------------------------------
+@given(
+    n_objects=st.integers(min_value=0, max_value=40),
+    limit=st.integers(min_value=1, max_value=9),
+)
+@settings(max_examples=40, deadline=None)
+def test_list_page_partitions_exactly(n_objects, limit):
+    """Any (store size, page size): pages concatenate to exactly the sorted
+    object set, every page shares one resourceVersion, and no page except
+    the last is short."""
+    from agac.kube.store import APIStore
+
+    store = APIStore()
+    for i in range(n_objects):
+        store.create(
+            corev1.Service(metadata=ObjectMeta(name=f"s-{i:03d}", namespace="default"))
+        )
+    seen, rvs, token, pages = [], [], None, 0
+    while True:
+        items, rv, token = store.list_page("Service", limit=limit, continue_token=token)
+        seen.extend(o.metadata.name for o in items)
+        rvs.append(rv)
+        pages += 1
+        if token is None:
+            break
+        assert len(items) == limit  # only the last page may be short
+        assert pages <= n_objects + 1  # termination
+    assert seen == [f"s-{i:03d}" for i in range(n_objects)]
+    assert len(set(rvs)) == 1
+
+
+@given(
+    n_objects=st.integers(min_value=3, max_value=25),
+    limit=st.integers(min_value=1, max_value=6),
+    churn_creates=st.lists(
+        st.integers(min_value=100, max_value=140), max_size=6, unique=True
+    ),
+    churn_deletes=st.integers(min_value=0, max_value=2),
+)
+@settings(max_examples=40, deadline=None)
+def test_list_page_under_concurrent_churn_never_duplicates(
+    n_objects, limit, churn_creates, churn_deletes
+):
+    """Mutations between pages may make the result miss brand-new objects
+    (real apiserver continuation semantics) but must never duplicate a
+    name or lose an object that existed before the list started and was
+    never deleted."""
+    from agac.kube.store import APIStore
+
+    store = APIStore()
+    initial = [f"s-{i:03d}" for i in range(n_objects)]
+    for name in initial:
+        store.create(corev1.Service(metadata=ObjectMeta(name=name, namespace="default")))
+    deleted = set()
+    seen, token = [], None
+    first = True
+    while True:
+        items, rv, token = store.list_page("Service", limit=limit, continue_token=token)
+        seen.extend(o.metadata.name for o in items)
+        if first:
+            # churn between the first and later pages
+            for i in churn_creates:
+                store.create(
+                    corev1.Service(
+                        metadata=ObjectMeta(name=f"s-{i:03d}", namespace="default")
+                    )
+                )
+            for name in initial[:churn_deletes]:
+                # deleting objects the scan ALREADY PASSED (page 1 covers
+                # the smallest names) keeps "never lose a survivor" exact
+                store.delete("Service", "default", name)
+                deleted.add(name)
+            first = False
+        if token is None:
+            break
+    assert len(seen) == len(set(seen))  # no duplicates, ever
+    survivors = [n for n in initial if n not in deleted]
+    assert set(survivors) <= set(seen)  # no pre-existing survivor lost
