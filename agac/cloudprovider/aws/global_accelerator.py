@@ -294,6 +294,11 @@ class GlobalAcceleratorMixin:
             },
         ):
             metrics.observe_hint("globalaccelerator", "hit")
+            # stash for same-reconcile reuse by _accelerator_changed (the
+            # reference re-lists tags it fetched moments earlier in the
+            # same sync, global_accelerator.go:87-110 then :412-437; the
+            # data is identical modulo a millisecond write race)
+            self._hint_verified_tags = (accelerator.accelerator_arn, tags)
             return [accelerator]
         metrics.observe_hint("globalaccelerator", "stale")
         return None
@@ -311,6 +316,10 @@ class GlobalAcceleratorMixin:
         port_changed,
         hint_arn: Optional[str] = None,
     ):
+        # one-reconcile scope for the hint-tag reuse (the controller builds
+        # a fresh AWS per reconcile like the reference's NewAWS, but guard
+        # direct library reuse of one instance across ensures)
+        self._hint_verified_tags = None
         lb = self.get_load_balancer(lb_name)
         if lb.dns_name != hostname:
             raise ValueError(f"LoadBalancer's DNS name is not matched: {lb.dns_name}")
@@ -462,11 +471,15 @@ class GlobalAcceleratorMixin:
             return True
         if accelerator.name != accelerator_name(resource, obj):
             return True
-        try:
-            tags = self._list_tags_for_accelerator(accelerator.accelerator_arn)
-        except Exception as e:
-            logger.warning("listing tags for %s failed: %s", accelerator.accelerator_arn, e)
-            return False
+        cached = getattr(self, "_hint_verified_tags", None)
+        if cached is not None and cached[0] == accelerator.accelerator_arn:
+            tags = cached[1]
+        else:
+            try:
+                tags = self._list_tags_for_accelerator(accelerator.accelerator_arn)
+            except Exception as e:
+                logger.warning("listing tags for %s failed: %s", accelerator.accelerator_arn, e)
+                return False
         target = {
             GLOBAL_ACCELERATOR_MANAGED_TAG_KEY: "true",
             GLOBAL_ACCELERATOR_OWNER_TAG_KEY: accelerator_owner_tag_value(
