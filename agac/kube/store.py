@@ -439,7 +439,10 @@ class APIStore:
         from .client import class_for_kind
 
         store = cls()
-        store._rv = itertools.count(int(snapshot.get("nextResourceVersion", 1)))
+        next_rv = int(snapshot.get("nextResourceVersion", 1))
+        store._rv = itertools.count(next_rv)
+        # keep latest_rv() (bookmark source) monotonic across a restore
+        store._last_rv = max(0, next_rv - 1)
         for entry in snapshot.get("objects", []):
             obj_cls = class_for_kind(entry["kind"])
             obj = metalib.from_dict(obj_cls, entry["object"])
