@@ -174,3 +174,28 @@ def test_route53_drift_repaired():
         wait_for(lambda: alias_dns() == good, "route53 drift repair", timeout=5.0)
     finally:
         stop.set()
+
+
+def test_cli_flag_maps_to_all_three_configs(monkeypatch):
+    """--cloud-resync-minutes reaches every controller config (in seconds)."""
+    from click.testing import CliRunner
+
+    from agac import cli as climod
+
+    captured = {}
+
+    class FakeManager:
+        def run(self, kube_client, config, cloud_factory, stop, **kw):
+            captured["config"] = config
+            stop.set()
+
+    monkeypatch.setattr("agac.manager.Manager", FakeManager)
+    result = CliRunner().invoke(climod.cli, [
+        "controller", "--api", "memory", "--no-leader-elect",
+        "--cloud-resync-minutes", "2.5",
+    ])
+    assert result.exit_code == 0, result.output
+    config = captured["config"]
+    assert config.global_accelerator.cloud_resync_period == 150.0
+    assert config.route53.cloud_resync_period == 150.0
+    assert config.endpoint_group_binding.cloud_resync_period == 150.0
