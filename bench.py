@@ -84,13 +84,21 @@ def build_stack(objects: int, workers: int, scenario: str = "ga", api: str = "me
     if not manager.wait_until_ready():
         raise RuntimeError("controllers did not become ready")
 
+    ext_groups = []
     if scenario == "full":
         backend.route53.create_hosted_zone("bench.example.com")
-        ext_acc = backend.ga.create_accelerator("external-bench")
-        ext_listener = backend.ga.create_listener(
-            ext_acc.accelerator_arn, [awstypes.PortRange(80, 80)], "TCP"
-        )
-        ext_group = backend.ga.create_endpoint_group(ext_listener.listener_arn, region)
+        # one external endpoint group per 8 bindings: real AWS caps an
+        # endpoint group at 10 endpoints, so piling every binding into a
+        # single group would be an unrealistic O(N^2) shape
+        n_bindings = max(1, objects // 4)
+        for g in range((n_bindings + 7) // 8):
+            ext_acc = backend.ga.create_accelerator(f"external-bench-{g}")
+            ext_listener = backend.ga.create_listener(
+                ext_acc.accelerator_arn, [awstypes.PortRange(80, 80)], "TCP"
+            )
+            ext_groups.append(
+                backend.ga.create_endpoint_group(ext_listener.listener_arn, region)
+            )
 
     services = []
     bindings = []
@@ -120,10 +128,11 @@ def build_stack(objects: int, workers: int, scenario: str = "ga", api: str = "me
 
     if scenario == "full":
         for i in range(max(1, objects // 4)):
+            group = ext_groups[i // 8]
             binding = egb.EndpointGroupBinding(
                 metadata=ObjectMeta(name=f"bind-{i}", namespace="default"),
                 spec=egb.EndpointGroupBindingSpec(
-                    endpoint_group_arn=ext_group.endpoint_group_arn,
+                    endpoint_group_arn=group.endpoint_group_arn,
                     weight=100,
                     service_ref=egb.ServiceReference(name=f"svc-{i}"),
                 ),
