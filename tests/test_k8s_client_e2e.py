@@ -196,3 +196,58 @@ def test_leader_election_over_production_client(env):
     finally:
         stop.set()
         t.join(timeout=5.0)
+
+
+def test_apiserver_bounce_with_persisted_state():
+    """The apiserver process dies mid-flight and comes back on the same
+    port from a dumped snapshot (agac apiserver --state-file semantics):
+    the controllers' informers must ride out the outage (connection
+    refused -> backoff -> relist), not duplicate existing cloud state, and
+    reconcile objects created after the restart."""
+    server = APIServer(APIStore(), watch_idle_seconds=0.1)
+    server.start()
+    port = server.port
+    client = K8sKubeClient(RestConfig(host=f"http://127.0.0.1:{port}"))
+    backend = FakeAWSBackend()
+    stop = threading.Event()
+    manager = Manager()
+    manager.run(client, ControllerConfig(), FakeCloudFactory(backend), stop,
+                resync_period=300.0, block=False)
+    try:
+        assert manager.wait_until_ready()
+        lb1 = backend.elbv2.create_load_balancer("bounce1", region=REGION)
+        client.create(managed_service("bounce1", lb1))
+        wait_for(lambda: len(backend.ga.list_accelerators()[0]) == 1, "initial")
+
+        snapshot = server.store.dump()
+        server.shutdown()
+        time.sleep(0.5)  # informers hit connection-refused and back off
+
+        server2 = APIServer(APIStore.load(snapshot), port=port,
+                            watch_idle_seconds=0.1)
+        server2.start()
+        try:
+            # new work created after the restart reconciles
+            lb2 = backend.elbv2.create_load_balancer("bounce2", region=REGION)
+            deadline = time.monotonic() + 20
+            while True:
+                try:
+                    client.create(managed_service("bounce2", lb2))
+                    break
+                except Exception:
+                    if time.monotonic() > deadline:
+                        raise
+                    time.sleep(0.1)
+            wait_for(lambda: len(backend.ga.list_accelerators()[0]) == 2,
+                     "post-restart reconcile", timeout=30.0)
+            # and the pre-restart object was NOT duplicated
+            names = sorted(a.name for a in backend.ga.list_accelerators()[0])
+            assert names == ["service-default-bounce1", "service-default-bounce2"]
+        finally:
+            server2.shutdown()
+    finally:
+        stop.set()
+        try:
+            server.shutdown()
+        except Exception:
+            pass
