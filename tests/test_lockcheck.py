@@ -11,12 +11,20 @@ from agac import lockcheck
 
 @pytest.fixture
 def checker():
+    # preserve any session-wide instrumentation (AGAC_LOCKCHECK=1 installs
+    # in conftest): snapshot the global graph, run this test isolated, then
+    # restore — so these self-tests never weaken the session-level check
+    was_installed = lockcheck._installed
+    prior_edges = dict(lockcheck._edges)
     lockcheck.reset()
     lockcheck.install()
     yield lockcheck
     lockcheck.check()  # note: tests that EXPECT a cycle reset before exit
-    lockcheck.uninstall()
+    if not was_installed:
+        lockcheck.uninstall()
     lockcheck.reset()
+    with lockcheck._graph_lock:
+        lockcheck._edges.update(prior_edges)
 
 
 def test_consistent_order_is_clean(checker):
