@@ -45,21 +45,36 @@ if _AVAILABLE:
     )
 
 
+# prometheus_client's .labels() re-validates and re-hashes on every call;
+# the label sets here are tiny and hot (every reconcile/AWS op), so memoize
+# the child metric objects
+_label_cache: dict = {}
+
+
+def _child(metric, *labels):
+    key = (id(metric), labels)
+    child = _label_cache.get(key)
+    if child is None:
+        child = _label_cache[key] = metric.labels(*labels)
+    return child
+
+
 def observe_reconcile(queue_name: str, outcome: str, seconds: float):
     if _AVAILABLE:
-        RECONCILE_TOTAL.labels(queue=queue_name or "unknown", outcome=outcome).inc()
-        RECONCILE_DURATION.labels(queue=queue_name or "unknown").observe(seconds)
+        q = queue_name or "unknown"
+        _child(RECONCILE_TOTAL, q, outcome).inc()
+        _child(RECONCILE_DURATION, q).observe(seconds)
 
 
 def observe_hint(controller: str, outcome: str):
     """outcome: hit | stale | error (stale = tags no longer match)."""
     if _AVAILABLE:
-        HINT_TOTAL.labels(controller=controller, outcome=outcome).inc()
+        _child(HINT_TOTAL, controller, outcome).inc()
 
 
 def observe_aws_call(service: str, operation: str):
     if _AVAILABLE:
-        AWS_API_CALLS.labels(service=service, operation=operation).inc()
+        _child(AWS_API_CALLS, service, operation).inc()
 
 
 def set_queue_depth(queue_name: str, depth: int):
