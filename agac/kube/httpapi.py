@@ -123,6 +123,8 @@ class _Handler(BaseHTTPRequestHandler):
                         int(rv) if rv else None,
                         k8s_style=True,
                         bookmarks=query.get("allowWatchBookmarks") in ("true", "1"),
+                        timeout_seconds=float(query["timeoutSeconds"])
+                        if query.get("timeoutSeconds") else None,
                     )
                     return True
                 limit = int(query["limit"]) if query.get("limit") else None
@@ -286,23 +288,35 @@ class _Handler(BaseHTTPRequestHandler):
             kind, query.get("namespace"), int(rv) if rv is not None else None,
             k8s_style=False,
             bookmarks=query.get("allowWatchBookmarks") in ("true", "1"),
+            timeout_seconds=float(query["timeoutSeconds"])
+            if query.get("timeoutSeconds") else None,
         )
 
     def _serve_watch_stream(self, kind, namespace, rv, k8s_style: bool,
-                            bookmarks: bool = False):
+                            bookmarks: bool = False,
+                            timeout_seconds=None):
         """Chunked ndjson event stream.  k8s_style frames events as the real
         apiserver does ({"type", "object"}); the native scheme adds a
         top-level resourceVersion.  With ``allowWatchBookmarks=true``, idle
         periods emit BOOKMARK events carrying only metadata.resourceVersion
         (the apiserver's watch-bookmark contract) so clients can advance
         their resume point without object traffic."""
+        import time as _time
+
         watch = self.store.watch(kind, namespace, rv)
+        deadline = (
+            _time.monotonic() + timeout_seconds if timeout_seconds else None
+        )
         self.send_response(200)
         self.send_header("Content-Type", "application/json")
         self.send_header("Transfer-Encoding", "chunked")
         self.end_headers()
         try:
             while True:
+                if deadline is not None and _time.monotonic() >= deadline:
+                    # ?timeoutSeconds= elapsed: end the stream like a real
+                    # apiserver; the client re-watches from its resume rv
+                    break
                 # Snapshot the rv BEFORE waiting: anything issued after this
                 # point is still in the watch queue, so bookmarking at this
                 # rv can never skip an undelivered event.

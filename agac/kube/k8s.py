@@ -115,7 +115,8 @@ class K8sKubeClient(KubeClient):
 
     def __init__(self, config: RestConfig, timeout: float = 10.0,
                  page_size: int = 500, max_retries: int = 5,
-                 field_manager: str = "aws-global-accelerator-controller"):
+                 field_manager: str = "aws-global-accelerator-controller",
+                 watch_timeout_seconds: float = 300.0):
         self.config = config
         self.base_url = config.host
         self.timeout = timeout
@@ -125,6 +126,10 @@ class K8sKubeClient(KubeClient):
         # tracking; sent as ?fieldManager= on every mutating verb, the
         # client-go behavior the reference inherits)
         self.field_manager = field_manager
+        # server-side watch deadline (client-go sends 5-10 min): the server
+        # cleanly ends the stream and the informer re-watches from its
+        # resume rv — keeps long-lived watches from going half-open
+        self.watch_timeout_seconds = watch_timeout_seconds
         self.session = requests.Session()
         if config.token:
             self.session.headers["Authorization"] = f"Bearer {config.token}"
@@ -276,6 +281,8 @@ class K8sKubeClient(KubeClient):
 
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
         params = {"watch": "true", "allowWatchBookmarks": "true"}
+        if self.watch_timeout_seconds:
+            params["timeoutSeconds"] = str(int(self.watch_timeout_seconds))
         if resource_version is not None:
             params["resourceVersion"] = str(resource_version)
         r = self.session.get(

@@ -364,3 +364,31 @@ class TestWatchReconnect:
         finally:
             stop.set()
             informer.stop()
+
+
+class TestWatchTimeout:
+    def test_server_honors_timeout_seconds_and_informer_survives(self, server):
+        """?timeoutSeconds= ends the stream like a real apiserver; the
+        informer reconnects across expiries without losing events."""
+        client = k8s_client(server, watch_timeout_seconds=1)
+        from agac.kube.informer import Informer, wait_for_cache_sync
+
+        stop = threading.Event()
+        informer = Informer(client, "Service", resync_period=0)
+        adds = []
+        informer.add_event_handler(on_add=lambda o: adds.append(o.metadata.name))
+        informer.run(stop)
+        try:
+            assert wait_for_cache_sync(stop, informer)
+            # run across several 1s watch expiries, creating during each
+            for i in range(3):
+                server.store.create(mk_service(f"tw-{i}"))
+                deadline = time.monotonic() + 10
+                while f"tw-{i}" not in adds:
+                    assert time.monotonic() < deadline
+                    time.sleep(0.02)
+                time.sleep(1.2)  # guarantee at least one expiry between
+            assert sorted(adds) == ["tw-0", "tw-1", "tw-2"]
+        finally:
+            stop.set()
+            informer.stop()
