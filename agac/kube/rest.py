@@ -122,9 +122,11 @@ class _RestWatch:
 
 
 class RestKubeClient(KubeClient):
-    def __init__(self, base_url: str, timeout: float = 10.0):
+    def __init__(self, base_url: str, timeout: float = 10.0,
+                 watch_timeout_seconds: float = 300.0):
         self.base_url = base_url.rstrip("/")
         self.timeout = timeout
+        self.watch_timeout_seconds = watch_timeout_seconds
         self.session = requests.Session()
 
     # -- raw verbs ---------------------------------------------------------
@@ -227,6 +229,8 @@ class RestKubeClient(KubeClient):
 
     def watch(self, kind: str, namespace: Optional[str] = None, resource_version=None):
         params = {"allowWatchBookmarks": "true"}
+        if self.watch_timeout_seconds:
+            params["timeoutSeconds"] = str(int(self.watch_timeout_seconds))
         if namespace:
             params["namespace"] = namespace
         if resource_version is not None:
