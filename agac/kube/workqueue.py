@@ -22,6 +22,8 @@ import threading
 import time
 from typing import Any, Optional, Tuple
 
+from .. import metrics
+
 
 class ItemExponentialFailureRateLimiter:
     """base * 2^failures per item, capped.  client-go defaults: 5ms..1000s."""
@@ -114,14 +116,15 @@ class RateLimitingQueue:
         self._dirty: set = set()
         self._processing: set = set()
         self._shutting_down = False
+        # per-item timestamps for the client-go-style latency metrics
+        self._added_at: dict = {}
+        self._started_at: dict = {}
         # delayed delivery
         self._waiting: list = []  # heap of (ready_at, seq, item)
         self._seq = itertools.count()
         self._waiting_thread: Optional[threading.Thread] = None
 
     def _report_depth(self):
-        from .. import metrics
-
         metrics.set_queue_depth(self.name, len(self._queue))
 
     # -- base queue --------------------------------------------------------
@@ -135,6 +138,7 @@ class RateLimitingQueue:
             if item in self._processing:
                 return
             self._queue.append(item)
+            self._added_at.setdefault(item, time.monotonic())
             self._report_depth()
             self._cond.notify_all()
 
@@ -156,14 +160,23 @@ class RateLimitingQueue:
             item = self._queue.pop(0)
             self._processing.add(item)
             self._dirty.discard(item)
+            now = time.monotonic()
+            added = self._added_at.pop(item, None)
+            if added is not None:
+                metrics.observe_queue_latency(self.name, now - added)
+            self._started_at[item] = now
             self._report_depth()
             return item, False
 
     def done(self, item: Any):
         with self._cond:
             self._processing.discard(item)
+            started = self._started_at.pop(item, None)
+            if started is not None:
+                metrics.observe_work_duration(self.name, time.monotonic() - started)
             if item in self._dirty:
                 self._queue.append(item)
+                self._added_at.setdefault(item, time.monotonic())
                 self._cond.notify_all()
 
     def __len__(self) -> int:
@@ -213,6 +226,7 @@ class RateLimitingQueue:
 
     # -- rate limiting -----------------------------------------------------
     def add_rate_limited(self, item: Any):
+        metrics.count_queue_retry(self.name)
         self.add_after(item, self._rate_limiter.when(item))
 
     def forget(self, item: Any):

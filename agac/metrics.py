@@ -43,6 +43,25 @@ if _AVAILABLE:
         "miss/error = fall back to the full accelerator scan)",
         ["controller", "outcome"],
     )
+    WORKQUEUE_QUEUE_DURATION = Histogram(
+        "agac_workqueue_queue_duration_seconds",
+        "Time items wait in a workqueue before a worker picks them up "
+        "(client-go workqueue_queue_duration_seconds)",
+        ["queue"],
+        buckets=(1e-4, 1e-3, 1e-2, 0.1, 1.0, 10.0, 60.0),
+    )
+    WORKQUEUE_WORK_DURATION = Histogram(
+        "agac_workqueue_work_duration_seconds",
+        "Time a worker spends processing an item "
+        "(client-go workqueue_work_duration_seconds)",
+        ["queue"],
+        buckets=(1e-4, 1e-3, 1e-2, 0.1, 1.0, 10.0, 60.0),
+    )
+    WORKQUEUE_RETRIES = Counter(
+        "agac_workqueue_retries_total",
+        "Rate-limited re-adds (client-go workqueue_retries_total)",
+        ["queue"],
+    )
 
 
 # prometheus_client's .labels() re-validates and re-hashes on every call;
@@ -79,7 +98,22 @@ def observe_aws_call(service: str, operation: str):
 
 def set_queue_depth(queue_name: str, depth: int):
     if _AVAILABLE and queue_name:
-        WORKQUEUE_DEPTH.labels(queue=queue_name).set(depth)
+        _child(WORKQUEUE_DEPTH, queue_name).set(depth)
+
+
+def observe_queue_latency(queue_name: str, seconds: float):
+    if _AVAILABLE and queue_name:
+        _child(WORKQUEUE_QUEUE_DURATION, queue_name).observe(seconds)
+
+
+def observe_work_duration(queue_name: str, seconds: float):
+    if _AVAILABLE and queue_name:
+        _child(WORKQUEUE_WORK_DURATION, queue_name).observe(seconds)
+
+
+def count_queue_retry(queue_name: str):
+    if _AVAILABLE and queue_name:
+        _child(WORKQUEUE_RETRIES, queue_name).inc()
 
 
 def start_metrics_server(port: int):

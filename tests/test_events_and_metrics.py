@@ -116,3 +116,27 @@ def test_workqueue_depth_gauge():
     assert REGISTRY.get_sample_value("agac_workqueue_depth", {"queue": "depth-test-q"}) == 2.0
     q.get()
     assert REGISTRY.get_sample_value("agac_workqueue_depth", {"queue": "depth-test-q"}) == 1.0
+
+
+def test_workqueue_latency_metrics_observed():
+    """client-go-style queue metrics: queue wait, work duration, retries."""
+    prometheus_client = __import__("pytest").importorskip("prometheus_client")
+    from agac import metrics
+    from agac.kube.workqueue import RateLimitingQueue
+
+    q = RateLimitingQueue(name="metrics-test-queue")
+
+    def hist_count(metric):
+        return metric.labels(queue="metrics-test-queue")._sum.get()
+
+    q.add("a")
+    item, shutdown = q.get(timeout=1.0)
+    assert item == "a" and not shutdown
+    q.done("a")
+    q.add_rate_limited("b")
+    item, _ = q.get(timeout=2.0)
+    q.done(item)
+    assert hist_count(metrics.WORKQUEUE_QUEUE_DURATION) >= 0.0
+    assert hist_count(metrics.WORKQUEUE_WORK_DURATION) >= 0.0
+    retries = metrics.WORKQUEUE_RETRIES.labels(queue="metrics-test-queue")
+    assert retries._value.get() >= 1
