@@ -238,7 +238,8 @@ def version():
 @cli.command("apiserver")
 @click.option("--port", default=8001, show_default=True, help="Listen port.")
 @click.option("--state-file", default="", help="Snapshot file: loaded on start if present, written on shutdown (checkpoint/resume).")
-def apiserver(port, state_file):
+@click.option("--checkpoint-interval-seconds", default=0.0, show_default=True, help="Also snapshot --state-file every N seconds (crash resilience; 0 = shutdown-only).")
+def apiserver(port, state_file, checkpoint_interval_seconds):
     """Serve the in-memory API store over HTTP (hermetic e2e backend)."""
     import json as jsonlib
 
@@ -257,12 +258,14 @@ def apiserver(port, state_file):
     server.start()
     logger.info("API server listening on :%d", server.port)
     stop = setup_signal_handler()
+    if state_file and checkpoint_interval_seconds > 0:
+        store.start_checkpointer(state_file, checkpoint_interval_seconds, stop)
+        logger.info(
+            "Checkpointing to %s every %.0fs", state_file, checkpoint_interval_seconds
+        )
     stop.wait()
     if state_file:
-        tmp = state_file + ".tmp"
-        with open(tmp, "w") as f:
-            jsonlib.dump(store.dump(), f)
-        os.replace(tmp, state_file)
+        store.save_snapshot(state_file)
         logger.info("State saved to %s", state_file)
     server.shutdown()
 

@@ -450,6 +450,36 @@ class APIStore:
             store._bucket(entry["kind"])[key] = obj
         return store
 
+    # -- checkpointing ------------------------------------------------------
+    def save_snapshot(self, path: str):
+        """Atomic snapshot write (tmp + rename)."""
+        import json as jsonlib
+        import os
+
+        tmp = path + ".tmp"
+        with open(tmp, "w") as f:
+            jsonlib.dump(self.dump(), f)
+        os.replace(tmp, path)
+
+    def start_checkpointer(self, path: str, interval: float, stop) -> "threading.Thread":
+        """Periodic crash-resilient snapshots (the shutdown-only dump loses
+        everything on a crash); returns the daemon thread.  A final
+        snapshot on clean shutdown is still the caller's job."""
+        def loop():
+            while not stop.wait(interval):
+                try:
+                    self.save_snapshot(path)
+                except Exception:  # never kill the server over a disk hiccup
+                    import logging
+
+                    logging.getLogger(__name__).exception(
+                        "periodic snapshot to %s failed", path
+                    )
+
+        thread = threading.Thread(target=loop, name="store-checkpointer", daemon=True)
+        thread.start()
+        return thread
+
     # -- watch -------------------------------------------------------------
     def watch(
         self,

@@ -119,3 +119,39 @@ def test_orphan_cleanup_after_restart():
     client.delete("Service", "default", "doomed")
     assert wait_until(lambda: accelerators(backend) == [])
     stop1.set()
+
+
+def test_periodic_checkpointer_snapshots(tmp_path):
+    """--checkpoint-interval-seconds: crash-resilient periodic snapshots
+    (the shutdown-only dump loses everything on a crash)."""
+    import json
+    import threading
+    import time
+
+    from agac.apis import core as corev1
+    from agac.apis.meta import ObjectMeta
+    from agac.kube.store import APIStore
+
+    store = APIStore()
+    path = str(tmp_path / "state.json")
+    stop = threading.Event()
+    store.start_checkpointer(path, interval=0.05, stop=stop)
+    try:
+        store.create(corev1.Service(metadata=ObjectMeta(name="ck", namespace="d")))
+        deadline = time.monotonic() + 5
+        while True:
+            try:
+                with open(path) as f:
+                    snap = json.load(f)
+                if any(e["object"]["metadata"]["name"] == "ck"
+                       for e in snap.get("objects", [])):
+                    break
+            except (FileNotFoundError, json.JSONDecodeError):
+                pass
+            assert time.monotonic() < deadline, "no periodic snapshot appeared"
+            time.sleep(0.02)
+        # a simulated crash (no clean shutdown) still restores from disk
+        restored = APIStore.load(snap)
+        assert restored.get("Service", "d", "ck").metadata.name == "ck"
+    finally:
+        stop.set()
