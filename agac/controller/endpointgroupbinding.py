@@ -92,6 +92,13 @@ class EndpointGroupBindingController:
 
     # -- run (reference egb/controller.go:101-187) ---------------------------
     def run(self, threadiness: int, stop: threading.Event):
+        try:
+            self._run(threadiness, stop)
+        finally:
+            self.workqueue.shut_down()
+            self.recorder.stop()
+
+    def _run(self, threadiness: int, stop: threading.Event):
         logger.info("Starting EndpointGroupBinding controller")
         if not wait_for_cache_sync(
             stop, self.binding_informer, self.service_informer, self.ingress_informer
@@ -107,7 +114,6 @@ class EndpointGroupBindingController:
             CONTROLLER_AGENT_NAME,
         )
         stop.wait()
-        self.workqueue.shut_down()
 
     def _run_worker(self):
         while self._process_next_work_item():

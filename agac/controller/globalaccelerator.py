@@ -135,6 +135,16 @@ class GlobalAcceleratorController:
 
     # -- run (reference ga/controller.go:195-230) ---------------------------
     def run(self, threadiness: int, stop: threading.Event):
+        try:
+            self._run(threadiness, stop)
+        finally:
+            # teardown on EVERY exit path (incl. shutdown-before-synced):
+            # queues release workers, the recorder's writer thread exits
+            self.service_queue.shut_down()
+            self.ingress_queue.shut_down()
+            self.recorder.stop()
+
+    def _run(self, threadiness: int, stop: threading.Event):
         logger.info("Starting GlobalAccelerator controller")
         if not wait_for_cache_sync(stop, self.service_informer, self.ingress_informer):
             if stop.is_set():
@@ -162,8 +172,6 @@ class GlobalAcceleratorController:
         )
         stop.wait()
         logger.info("Shutting down workers")
-        self.service_queue.shut_down()
-        self.ingress_queue.shut_down()
 
     def _run_service_worker(self):
         while reconcile.process_next_work_item(

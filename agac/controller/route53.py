@@ -144,6 +144,14 @@ class Route53Controller:
 
     # -- run ----------------------------------------------------------------
     def run(self, threadiness: int, stop: threading.Event):
+        try:
+            self._run(threadiness, stop)
+        finally:
+            self.service_queue.shut_down()
+            self.ingress_queue.shut_down()
+            self.recorder.stop()
+
+    def _run(self, threadiness: int, stop: threading.Event):
         logger.info("Starting Route53 controller")
         if not wait_for_cache_sync(stop, self.service_informer, self.ingress_informer):
             if stop.is_set():
@@ -169,8 +177,6 @@ class Route53Controller:
             CONTROLLER_AGENT_NAME,
         )
         stop.wait()
-        self.service_queue.shut_down()
-        self.ingress_queue.shut_down()
 
     def _run_service_worker(self):
         while reconcile.process_next_work_item(

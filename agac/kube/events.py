@@ -61,6 +61,9 @@ class EventRecorder:
     def _writer(self):
         while True:
             item = self._queue.get()
+            if item is None:  # stop() sentinel
+                self._queue.task_done()
+                return
             try:
                 self._write(*item)
             except Exception:
@@ -102,6 +105,15 @@ class EventRecorder:
                 return True
             time.sleep(0.005)
         return self._queue.unfinished_tasks == 0
+
+    def stop(self):
+        """Terminate the writer thread after draining queued events
+        (controllers call this at shutdown so repeated manager lifecycles
+        — e.g. leadership churn — don't accumulate threads)."""
+        try:
+            self._queue.put_nowait(None)
+        except Exception:
+            pass
 
     def eventf(self, obj, event_type: str, reason: str, fmt: str, *args):
         self.event(obj, event_type, reason, fmt % args if args else fmt)

@@ -140,3 +140,41 @@ def test_workqueue_latency_metrics_observed():
     assert hist_count(metrics.WORKQUEUE_WORK_DURATION) >= 0.0
     retries = metrics.WORKQUEUE_RETRIES.labels(queue="metrics-test-queue")
     assert retries._value.get() >= 1
+
+
+def test_recorder_threads_do_not_leak_across_manager_lifecycles():
+    """Leadership churn restarts the manager; recorder writer threads must
+    terminate with their controller instead of accumulating."""
+    import threading
+    import time
+
+    from agac.cloudprovider.aws.client import FakeCloudFactory
+    from agac.cloudprovider.fake import FakeAWSBackend
+    from agac.manager import ControllerConfig, Manager
+
+    def cycle():
+        client = InMemoryKubeClient()
+        stop = threading.Event()
+        manager = Manager()
+        manager.run(client, ControllerConfig(), FakeCloudFactory(FakeAWSBackend()),
+                    stop, resync_period=300.0, block=False)
+        assert manager.wait_until_ready()
+        stop.set()
+        time.sleep(0.05)
+
+    cycle()  # warm-up lifecycle (imports, etc.)
+    time.sleep(0.3)
+    baseline = threading.active_count()
+    for _ in range(4):
+        cycle()
+    deadline = time.monotonic() + 10
+    while threading.active_count() > baseline + 6:
+        if time.monotonic() > deadline:
+            break
+        time.sleep(0.05)
+    # workers exit on queue shutdown, informer loops on stop, recorder
+    # writers on the sentinel: 4 extra lifecycles must not pile up the
+    # ~12 threads each one spawns
+    assert threading.active_count() <= baseline + 6, (
+        f"thread leak: baseline {baseline}, now {threading.active_count()}"
+    )
