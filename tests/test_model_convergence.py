@@ -259,9 +259,14 @@ def run_walk(seed: int, n_services: int = 6, n_ops: int = 60, api: str = "memory
             server.shutdown()
 
 
-@pytest.mark.parametrize("seed", [7, 23, 1009])
+@pytest.mark.parametrize("seed", [7, 23, 1009, 31337, 77, 5150])
 def test_random_walk_converges_to_model(seed):
     run_walk(seed)
+
+
+def test_long_walk_converges_to_model():
+    """A deeper walk: more services, more ops, more interleavings."""
+    run_walk(999331, n_services=12, n_ops=200)
 
 
 def test_random_walk_over_production_wire_client():
