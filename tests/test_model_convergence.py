@@ -273,3 +273,122 @@ def test_random_walk_over_production_wire_client():
     """The same walk through K8sKubeClient -> HTTP (page_size=3 keeps the
     informer syncs multi-page)."""
     run_walk(4242, api="http")
+
+
+# ---------------------------------------------------------------------------
+# Ingress walk: the ALB-ingress paths through the same model audit
+# ---------------------------------------------------------------------------
+LISTEN_PORTS = "alb.ingress.kubernetes.io/listen-ports"
+
+
+def run_ingress_walk(seed: int, n_ingresses: int = 5, n_ops: int = 50):
+    rng = random.Random(seed)
+    backend = FakeAWSBackend()
+    client = InMemoryKubeClient()
+    stop = threading.Event()
+    manager = Manager()
+    factory = FakeCloudFactory(backend, ga_missing_retry=0.1)
+    manager.run(client, ControllerConfig(), factory, stop,
+                resync_period=300.0, block=False)
+    assert manager.wait_until_ready()
+    model = {}  # name -> {"managed": bool, "ports": [..]}
+    lbs = {}
+
+    def k8s_ingress(name):
+        m = model[name]
+        annotations = {LISTEN_PORTS: str(
+            [{"HTTP": p} for p in m["ports"]]).replace("'", '"')}
+        if m["managed"]:
+            annotations[MANAGED] = "true"
+        return corev1.Ingress(
+            metadata=ObjectMeta(name=name, namespace="default",
+                                annotations=annotations),
+            spec=corev1.IngressSpec(ingress_class_name="alb"),
+            status=corev1.IngressStatus(
+                load_balancer=corev1.IngressLoadBalancerStatus(
+                    ingress=[corev1.IngressLoadBalancerIngress(
+                        hostname=lbs[name].dns_name)]
+                )
+            ),
+        )
+
+    def push(name):
+        desired = k8s_ingress(name)
+        for _ in range(20):
+            try:
+                live = client.get("Ingress", "default", name)
+            except NotFoundError:
+                client.create(desired)
+                return
+            live.metadata.annotations = desired.metadata.annotations
+            try:
+                client.update(live)
+                return
+            except Exception:
+                continue
+
+    try:
+        for _ in range(n_ops):
+            name = f"ing-{rng.randrange(n_ingresses)}"
+            op = rng.choice(["create", "manage", "unmanage", "ports", "delete"])
+            if name not in model:
+                if op == "delete":
+                    continue
+                if name not in lbs:
+                    # ALB hostname shape so the parser takes the ALB branch
+                    lbs[name] = backend.elbv2.create_load_balancer(
+                        name, region=REGION, lb_type="application",
+                    )
+                model[name] = {"managed": False, "ports": [80]}
+                push(name)
+                continue
+            if op == "create":
+                continue
+            if op == "manage":
+                model[name]["managed"] = True
+            elif op == "unmanage":
+                model[name]["managed"] = False
+            elif op == "ports":
+                model[name]["ports"] = sorted(
+                    rng.sample([80, 443, 8080, 8443, 9090], rng.randint(1, 3))
+                )
+            elif op == "delete":
+                client.delete("Ingress", "default", name)
+                del model[name]
+                continue
+            push(name)
+
+        def audit():
+            managed = {n: m for n, m in model.items() if m["managed"]}
+            accs, _ = backend.ga.list_accelerators()
+            owned = {}
+            for a in accs:
+                tags = {t.key: t.value for t in
+                        backend.ga.list_tags_for_resource(a.accelerator_arn)}
+                owner = tags.get(OWNER_TAG)
+                if owner is None:
+                    return False
+                owned.setdefault(owner, []).append(a)
+            if set(owned) != {f"ingress/default/{n}" for n in managed}:
+                return False
+            for name, m in managed.items():
+                entries = owned[f"ingress/default/{name}"]
+                if len(entries) != 1:
+                    return False
+                listeners, _ = backend.ga.list_listeners(entries[0].accelerator_arn)
+                if len(listeners) != 1:
+                    return False
+                if listeners[0].protocol != "TCP":  # ALB listeners are TCP
+                    return False
+                if sorted(p.from_port for p in listeners[0].port_ranges) != m["ports"]:
+                    return False
+            return True
+
+        settle(audit, what=f"ingress walk seed {seed}")
+    finally:
+        stop.set()
+
+
+@pytest.mark.parametrize("seed", [11, 4242, 90210])
+def test_ingress_random_walk_converges(seed):
+    run_ingress_walk(seed)
