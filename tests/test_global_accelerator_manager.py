@@ -407,3 +407,37 @@ class TestUserTagDrift:
         )
         tags = {x.key: x.value for x in backend.ga.list_tags_for_resource(arn)}
         assert tags.get("env") == "prod"  # stale but reference-faithful
+
+
+def test_multi_hostname_last_wins_parity():
+    """PARITY quirk 8b: two LB hostnames on one service converge to ONE
+    accelerator tagged with the LAST hostname processed (each ensure's
+    owner-scan update loop retags whatever the previous hostname's pass
+    ensured — reference global_accelerator.go:130-141)."""
+    from agac.apis import core as corev1
+    from agac.apis.meta import ObjectMeta
+    from agac.cloudprovider.aws.client import FakeCloudFactory
+    from agac.cloudprovider.fake import FakeAWSBackend
+
+    backend = FakeAWSBackend()
+    cloud = FakeCloudFactory(backend)("us-east-1")
+    lb_a = backend.elbv2.create_load_balancer("mh-a", region="us-east-1")
+    lb_b = backend.elbv2.create_load_balancer("mh-b", region="us-east-1")
+    svc = corev1.Service(
+        metadata=ObjectMeta(name="multi", namespace="default"),
+        spec=corev1.ServiceSpec(
+            type="LoadBalancer",
+            ports=[corev1.ServicePort(port=80, protocol="TCP")],
+        ),
+    )
+    for _ in range(3):  # stable across reconciles, not oscillating
+        for lb, name in ((lb_a, "mh-a"), (lb_b, "mh-b")):
+            cloud.ensure_global_accelerator_for_service(
+                svc, corev1.LoadBalancerIngress(hostname=lb.dns_name),
+                "c", name, "us-east-1",
+            )
+        accs, _ = backend.ga.list_accelerators()
+        assert len(accs) == 1
+        tags = {t.key: t.value for t in
+                backend.ga.list_tags_for_resource(accs[0].accelerator_arn)}
+        assert tags["aws-global-accelerator-target-hostname"] == lb_b.dns_name
