@@ -392,3 +392,54 @@ class TestWatchTimeout:
         finally:
             stop.set()
             informer.stop()
+
+
+class TestPagedSyncUnderChurn:
+    def test_informer_converges_while_store_churns(self, server):
+        """A multi-page initial sync races live creates/deletes; whatever
+        mix of continue-tokens, 410 restarts and watch replays happens, the
+        informer must end exactly consistent with the store."""
+        for i in range(900):
+            server.store.create(mk_service(f"base-{i:04d}"))
+        churn_stop = threading.Event()
+
+        def churner():
+            i = 0
+            while not churn_stop.is_set():
+                server.store.create(mk_service(f"churn-{i:04d}"))
+                if i >= 5:
+                    server.store.delete("Service", "default", f"churn-{i-5:04d}")
+                i += 1
+                time.sleep(0.002)
+
+        t = threading.Thread(target=churner, daemon=True)
+        client = k8s_client(server, page_size=50)  # 18+ pages
+        from agac.kube.informer import Informer, wait_for_cache_sync
+
+        stop = threading.Event()
+        informer = Informer(client, "Service", resync_period=0)
+        t.start()
+        informer.run(stop)
+        try:
+            assert wait_for_cache_sync(stop, informer, timeout=60.0)
+            churn_stop.set()
+            time.sleep(0.3)  # let the tail of the churn drain through watch
+            store_names = {
+                o.metadata.name for o in server.store.list("Service")[0]
+            }
+            deadline = time.monotonic() + 10
+            while True:
+                cache_names = {
+                    o.metadata.name for o in informer.cache_list()
+                }
+                if cache_names == store_names:
+                    break
+                assert time.monotonic() < deadline, (
+                    f"cache != store: missing={sorted(store_names - cache_names)[:5]} "
+                    f"extra={sorted(cache_names - store_names)[:5]}"
+                )
+                time.sleep(0.05)
+        finally:
+            churn_stop.set()
+            stop.set()
+            informer.stop()
