@@ -11,14 +11,16 @@ test-fast:
 
 # threading stress profile: repeat the concurrency-heavy suites
 test-race:
-	# lock-order checking (agac/lockcheck.py): the concurrency suites run
-	# with every project lock instrumented; a lock-order cycle (potential
-	# deadlock) fails the session — plus 3x stress repetition for flake
-	# surfacing (the go-test-race analogue this pure-Python tier can have)
-	for i in 1 2 3; do \
+	# lock-order checking (agac/lockcheck.py): every project lock is
+	# instrumented; a lock-order cycle (potential deadlock) fails the
+	# session (the go-test-race analogue this pure-Python tier can have).
+	# One full-suite pass + 2x stress repetition for flake surfacing.
+	AGAC_LOCKCHECK=1 $(PYTHON) -m pytest tests/ -q -m "not gpu"
+	for i in 1 2; do \
 		AGAC_LOCKCHECK=1 $(PYTHON) -m pytest tests/test_stress_concurrency.py \
 			tests/test_chaos_recovery.py tests/test_store_event_sourcing.py \
-			tests/test_fault_injection.py tests/test_cloud_resync.py -q || exit 1; \
+			tests/test_fault_injection.py tests/test_cloud_resync.py \
+			tests/test_failover_under_churn.py -q || exit 1; \
 	done
 
 bench:

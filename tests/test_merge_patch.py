@@ -120,3 +120,38 @@ class TestPatchOverTheWire:
         self.seed(client)
         client.services("d").patch("s", {"metadata": {"annotations": {"x": "y"}}})
         assert client.get("Service", "d", "s").metadata.annotations == {"x": "y"}
+
+
+class TestMergePatchProperties:
+    """RFC 7386 algebraic properties, fuzzed with hypothesis."""
+
+    def test_idempotence_and_null_removal(self):
+        from hypothesis import given, settings
+        from hypothesis import strategies as st
+
+        from agac.kube.patch import json_merge_patch
+
+        scalars = st.one_of(st.none(), st.booleans(), st.integers(),
+                            st.text(max_size=8))
+        docs = st.recursive(
+            scalars,
+            lambda children: st.dictionaries(
+                st.text(max_size=5), children, max_size=4
+            ),
+            max_leaves=12,
+        )
+
+        @given(doc=docs, patch=docs)
+        @settings(max_examples=200, deadline=None)
+        def check(doc, patch):
+            once = json_merge_patch(doc, patch)
+            twice = json_merge_patch(once, patch)
+            assert once == twice  # idempotent (RFC 7386 §2)
+            if isinstance(patch, dict) and isinstance(once, dict):
+                for k, v in patch.items():
+                    if v is None:
+                        assert k not in once  # null removes
+                    elif not isinstance(v, dict):
+                        assert once.get(k) == v  # scalar replaces
+
+        check()
