@@ -116,9 +116,14 @@ def main():
             except (ConflictError, ConnectionError, OSError):
                 time.sleep(0.01)
 
+    churn_pause = threading.Event()
+
     def churner():
         i = 0
         while not churn_stop.is_set():
+            if churn_pause.is_set():
+                time.sleep(0.02)
+                continue
             name = f"fs-{i % args.objects}"
             push(name, 8000 + (i % 500))
             i += 1
@@ -176,7 +181,7 @@ def main():
                                   "error": "standby never led"}))
                 break
             # quiesce churn briefly to audit an exact state
-            churn_stop.set()
+            churn_pause.set()
             time.sleep(0.3)
             audit_deadline = time.monotonic() + 60
             ok = False
@@ -199,9 +204,8 @@ def main():
                 "kill_to_converged_s": round(elapsed, 2), "converged": ok,
             }))
             sys.stdout.flush()
-            # resume churn with a fresh standby
-            churn_stop.clear()
-            threading.Thread(target=churner, daemon=True).start()
+            # resume churn with a fresh standby (single long-lived churner)
+            churn_pause.clear()
             active, standby = other, new_replica()
     finally:
         churn_stop.set()
