@@ -72,19 +72,35 @@ def http_admission(
     url: str,
     timeout: float = 10.0,
     verify=True,
+    failure_policy: str = "Fail",
 ) -> AdmissionWebhook:
-    """HTTP transport: POST the AdmissionReview to ``url``."""
+    """HTTP transport: POST the AdmissionReview to ``url``.
+
+    ``failure_policy`` mirrors ValidatingWebhookConfiguration: when the
+    webhook is unreachable/broken, ``Fail`` (the k8s default) rejects the
+    write, ``Ignore`` admits it."""
     import requests
 
     def review_fn(review: dict) -> dict:
-        response = requests.post(
-            url,
-            json=review,
-            headers={"Content-Type": "application/json"},
-            timeout=timeout,
-            verify=verify,
-        )
-        response.raise_for_status()
-        return response.json()
+        try:
+            response = requests.post(
+                url,
+                json=review,
+                headers={"Content-Type": "application/json"},
+                timeout=timeout,
+                verify=verify,
+            )
+            response.raise_for_status()
+            return response.json()
+        except Exception as e:
+            if failure_policy == "Ignore":
+                logger.warning(
+                    "admission webhook %s unreachable (%s); failurePolicy="
+                    "Ignore admits the request", url, e,
+                )
+                return {"response": {"allowed": True}}
+            raise AdmissionDeniedError(
+                f"failed calling webhook {url} (failurePolicy=Fail): {e}"
+            ) from e
 
     return AdmissionWebhook(kinds, operations, review_fn)

@@ -97,3 +97,45 @@ class TestBatchedWeightSync:
         desc = backend.ga.describe_endpoint_group(group.endpoint_group_arn)
         weights = {d.endpoint_id: d.weight for d in desc.endpoint_descriptions}
         assert weights == {"arn:lb1": 50, "arn:lb2": 50, "arn:other": 3}
+
+
+class TestAdmissionFailurePolicy:
+    """ValidatingWebhookConfiguration failurePolicy parity: an unreachable
+    webhook rejects writes under Fail (the k8s default) and admits under
+    Ignore."""
+
+    def _store_with_hook(self, failure_policy):
+        from agac.apis import endpointgroupbinding as egb
+        from agac.kube.admission import http_admission
+        from agac.kube.store import APIStore
+
+        store = APIStore()
+        store.admission_webhooks.append(http_admission(
+            kinds=["EndpointGroupBinding"],
+            operations=["CREATE", "UPDATE"],
+            url="http://127.0.0.1:1",  # nothing listens here
+            timeout=0.2,
+            failure_policy=failure_policy,
+        ))
+        from agac.apis.meta import ObjectMeta
+
+        binding = egb.EndpointGroupBinding(
+            metadata=ObjectMeta(name="fp", namespace="d"),
+            spec=egb.EndpointGroupBindingSpec(
+                endpoint_group_arn="arn:aws:globalaccelerator::1:x"),
+        )
+        return store, binding
+
+    def test_fail_policy_rejects_when_webhook_down(self):
+        import pytest
+
+        from agac.kube.admission import AdmissionDeniedError
+
+        store, binding = self._store_with_hook("Fail")
+        with pytest.raises(AdmissionDeniedError, match="failurePolicy=Fail"):
+            store.create(binding)
+
+    def test_ignore_policy_admits_when_webhook_down(self):
+        store, binding = self._store_with_hook("Ignore")
+        created = store.create(binding)
+        assert created.metadata.name == "fp"
