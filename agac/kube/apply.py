@@ -37,7 +37,10 @@ def apply_document(client, doc: dict) -> Tuple[str, str]:
         return "skipped", f"{api_version}/{kind}"
     cls = class_for_kind(kind)
     obj = from_dict(cls, doc)
-    namespace = obj.metadata.namespace or "default"
+    if gvr.namespaced:
+        namespace = obj.metadata.namespace or "default"
+    else:
+        namespace = ""  # cluster-scoped
     obj.metadata.namespace = namespace
     ident = f"{kind.lower()}/{namespace}/{obj.metadata.name}"
     try:

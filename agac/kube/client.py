@@ -11,6 +11,7 @@ from __future__ import annotations
 
 from typing import Optional
 
+from ..apis import admissionregistration as admissionv1
 from ..apis import core as corev1
 from ..apis import endpointgroupbinding as egbv1alpha1
 from .store import APIStore
@@ -21,6 +22,7 @@ _KINDS = {
     "Event": corev1.Event,
     "Lease": corev1.Lease,
     "EndpointGroupBinding": egbv1alpha1.EndpointGroupBinding,
+    "ValidatingWebhookConfiguration": admissionv1.ValidatingWebhookConfiguration,
 }
 
 

@@ -62,6 +62,10 @@ GVRS = [
     GroupVersionResource(
         "operator.h3poteto.dev", "v1alpha1", "endpointgroupbindings", "EndpointGroupBinding"
     ),
+    GroupVersionResource(
+        "admissionregistration.k8s.io", "v1", "validatingwebhookconfigurations",
+        "ValidatingWebhookConfiguration", namespaced=False,
+    ),
 ]
 
 BY_KIND = {g.kind: g for g in GVRS}

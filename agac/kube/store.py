@@ -142,18 +142,38 @@ class APIStore:
         self._event_log: deque = deque(maxlen=_EVENT_LOG_SIZE)
         # registered admission webhooks (agac.kube.admission), consulted
         # before create/update/delete commits — the apiserver side of
-        # ValidatingWebhookConfiguration
+        # ValidatingWebhookConfiguration.  VWC OBJECTS applied to the store
+        # are consulted too (dynamic admission); service references resolve
+        # through webhook_service_resolver (None = unresolvable, which the
+        # failurePolicy then governs).
         self.admission_webhooks: List = []
+        self.webhook_service_resolver = None
+
+    def _admission_active(self) -> bool:
+        return bool(self.admission_webhooks) or bool(
+            self._objects.get("ValidatingWebhookConfiguration")
+        )
 
     def _admit(self, kind: str, operation: str, old, new):
         """Run admission outside the store lock (webhook calls may do
-        network I/O); commit-time rv checks still serialize writers."""
-        if not self.admission_webhooks:
-            return
-        old_d = metalib.to_dict(old) if old is not None else None
-        new_d = metalib.to_dict(new) if new is not None else None
-        for hook in self.admission_webhooks:
-            hook.admit(kind, operation, old_d, new_d)
+        network I/O); commit-time rv checks still serialize writers.
+        Consults both programmatically registered hooks and any
+        ValidatingWebhookConfiguration OBJECTS in the store (dynamic
+        admission, like a real apiserver — agac/kube/dynamicadmission.py)."""
+        old_d = new_d = None
+        if self.admission_webhooks:
+            old_d = metalib.to_dict(old) if old is not None else None
+            new_d = metalib.to_dict(new) if new is not None else None
+            for hook in self.admission_webhooks:
+                hook.admit(kind, operation, old_d, new_d)
+        if self._bucket("ValidatingWebhookConfiguration"):
+            from . import dynamicadmission
+
+            if old_d is None and old is not None:
+                old_d = metalib.to_dict(old)
+            if new_d is None and new is not None:
+                new_d = metalib.to_dict(new)
+            dynamicadmission.admit(self, kind, operation, old_d, new_d)
 
     # -- helpers -----------------------------------------------------------
     def _bucket(self, kind: str) -> Dict[Tuple[str, str], object]:
@@ -304,7 +324,7 @@ class APIStore:
         from .validation import validate_object
 
         validate_object(obj)
-        if self.admission_webhooks:
+        if self._admission_active():
             try:
                 current = self.get(kind, obj.metadata.namespace, obj.metadata.name)
             except NotFoundError:
@@ -388,7 +408,7 @@ class APIStore:
 
     def delete(self, kind: str, namespace: str, name: str):
         """Finalizer-aware delete (kube-apiserver graceful deletion)."""
-        if self.admission_webhooks:
+        if self._admission_active():
             try:
                 current = self.get(kind, namespace, name)
             except NotFoundError:

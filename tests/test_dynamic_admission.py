@@ -1,0 +1,121 @@
+"""Dynamic admission: ValidatingWebhookConfiguration consumed as an API
+OBJECT (the real cluster mechanism, and what the reference's kind e2e
+exercises at e2e/e2e_test.go:77-103): apply config/webhook/manifests.yaml,
+resolve the service reference to a live `agac webhook` TLS server with the
+caBundle doing the trust, and the store enforces ARN immutability on
+writes — no programmatic hook registration anywhere."""
+
+import base64
+import subprocess
+
+import pytest
+
+from agac.apis import endpointgroupbinding as egb
+from agac.apis.meta import ObjectMeta
+from agac.kube.admission import AdmissionDeniedError
+from agac.kube.apply import apply_yaml
+from agac.kube.client import InMemoryKubeClient
+from agac.kube.store import APIStore
+from agac.webhook.server import WebhookServer
+
+
+@pytest.fixture(scope="module")
+def tls_webhook(tmp_path_factory):
+    d = tmp_path_factory.mktemp("dyncerts")
+    cert, key = d / "tls.crt", d / "tls.key"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(key), "-out", str(cert), "-days", "1",
+         "-subj", "/CN=127.0.0.1",
+         "-addext", "subjectAltName=IP:127.0.0.1"],
+        check=True, capture_output=True,
+    )
+    server = WebhookServer(port=0, tls_cert_file=str(cert), tls_key_file=str(key))
+    server.start()
+    yield server, cert.read_bytes()
+    server.shutdown()
+
+
+def binding(name="dyn", arn="arn:aws:globalaccelerator::1:x", weight=10):
+    return egb.EndpointGroupBinding(
+        metadata=ObjectMeta(name=name, namespace="default"),
+        spec=egb.EndpointGroupBindingSpec(endpoint_group_arn=arn, weight=weight),
+    )
+
+
+@pytest.fixture
+def store(tls_webhook):
+    server, ca_pem = tls_webhook
+    store = APIStore()
+    client = InMemoryKubeClient(store)
+    manifest = open("config/webhook/manifests.yaml").read()
+    # inject the caBundle the way cert-manager would on a live cluster
+    manifest = manifest.replace(
+        "  clientConfig:",
+        "  clientConfig:\n    caBundle: " + base64.b64encode(ca_pem).decode(),
+    )
+    results = apply_yaml(client, manifest)
+    assert results[0][0] == "created"
+
+    # service-reference resolution (clusters use <name>.<ns>.svc DNS)
+    def resolver(service_ref):
+        assert service_ref.name == "aws-global-accelerator-controller-webhook"
+        return f"https://127.0.0.1:{server.port}"
+
+    store.webhook_service_resolver = resolver
+    return store
+
+
+def test_arn_update_rejected_via_applied_manifest(store):
+    store.create(binding())
+    live = store.get("EndpointGroupBinding", "default", "dyn")
+    live.spec.endpoint_group_arn = "arn:changed"
+    with pytest.raises(AdmissionDeniedError, match="immutable"):
+        store.update(live)
+
+
+def test_weight_update_allowed(store):
+    store.create(binding(name="dyn2"))
+    live = store.get("EndpointGroupBinding", "default", "dyn2")
+    live.spec.weight = 99
+    updated = store.update(live)
+    assert updated.spec.weight == 99
+
+
+def test_create_passes_validator(store):
+    # CREATE is registered (ADVICE r1 fix) and the validator allows it
+    created = store.create(binding(name="dyn3"))
+    assert created.metadata.name == "dyn3"
+
+
+def test_unmatched_kind_skips_webhook(store):
+    from agac.apis import core as corev1
+
+    store.create(corev1.Service(metadata=ObjectMeta(name="svc", namespace="default")))
+
+
+def test_unresolvable_service_honors_failure_policy(tls_webhook):
+    server, ca_pem = tls_webhook
+    store = APIStore()
+    client = InMemoryKubeClient(store)
+    manifest = open("config/webhook/manifests.yaml").read()
+    apply_yaml(client, manifest)
+    # no resolver installed → service unresolvable → failurePolicy: Fail
+    with pytest.raises(AdmissionDeniedError, match="failurePolicy=Fail"):
+        store.create(binding(name="fp"))
+    # flip the config to Ignore via the API (like kubectl patch would)
+    vwc = store.get("ValidatingWebhookConfiguration", "",
+                    "aws-global-accelerator-controller-validating-webhook")
+    vwc.webhooks[0].failure_policy = "Ignore"
+    store.update(vwc)
+    created = store.create(binding(name="fp"))
+    assert created.metadata.name == "fp"
+
+
+def test_deleting_the_configuration_disables_admission(store):
+    store.delete("ValidatingWebhookConfiguration", "",
+                 "aws-global-accelerator-controller-validating-webhook")
+    store.create(binding(name="dyn4"))
+    live = store.get("EndpointGroupBinding", "default", "dyn4")
+    live.spec.endpoint_group_arn = "arn:changed-freely"
+    store.update(live)  # no webhook left to veto
