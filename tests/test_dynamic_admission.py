@@ -119,3 +119,34 @@ def test_deleting_the_configuration_disables_admission(store):
     live = store.get("EndpointGroupBinding", "default", "dyn4")
     live.spec.endpoint_group_arn = "arn:changed-freely"
     store.update(live)  # no webhook left to veto
+
+
+def test_vwc_over_the_wire_activates_admission():
+    """Cluster-scoped VWC round-trips the k8s REST surface and admission
+    applies to subsequent wire writes (no resolver installed here, so the
+    Fail policy rejects — proving the config is live server-side)."""
+    from agac.kube.httpapi import APIServer
+    from agac.kube.k8s import K8sKubeClient
+    from agac.kube.kubeconfig import RestConfig
+
+    server = APIServer(APIStore())
+    server.start()
+    try:
+        client = K8sKubeClient(RestConfig(host=server.url))
+        results = apply_yaml(client, open("config/webhook/manifests.yaml").read())
+        assert results[0][0] == "created"
+        # cluster-scoped read back over the wire
+        vwc = client.get("ValidatingWebhookConfiguration", "",
+                         "aws-global-accelerator-controller-validating-webhook")
+        assert vwc.webhooks[0].failure_policy == "Fail"
+        # admission is active server-side for wire writes
+        import pytest as _pytest
+
+        with _pytest.raises(AdmissionDeniedError, match="failurePolicy=Fail"):
+            client.create(binding(name="wire-fp"))
+        # unmatched kinds unaffected
+        from agac.apis import core as corev1
+
+        client.create(corev1.Service(metadata=ObjectMeta(name="ok", namespace="default")))
+    finally:
+        server.shutdown()
