@@ -239,7 +239,8 @@ def version():
 @click.option("--port", default=8001, show_default=True, help="Listen port.")
 @click.option("--state-file", default="", help="Snapshot file: loaded on start if present, written on shutdown (checkpoint/resume).")
 @click.option("--checkpoint-interval-seconds", default=0.0, show_default=True, help="Also snapshot --state-file every N seconds (crash resilience; 0 = shutdown-only).")
-def apiserver(port, state_file, checkpoint_interval_seconds):
+@click.option("--resolve-webhook-service", "webhook_services", multiple=True, metavar="NAME.NAMESPACE=URL", help="Resolve a ValidatingWebhookConfiguration service reference to a URL (clusters use <name>.<ns>.svc DNS; repeatable).")
+def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services):
     """Serve the in-memory API store over HTTP (hermetic e2e backend)."""
     import json as jsonlib
 
@@ -253,6 +254,22 @@ def apiserver(port, state_file, checkpoint_interval_seconds):
         logger.info("Restored state from %s", state_file)
     else:
         store = APIStore()
+
+    if webhook_services:
+        mapping = {}
+        for entry in webhook_services:
+            ref, _, url = entry.partition("=")
+            if not url:
+                raise click.UsageError(
+                    f"--resolve-webhook-service wants NAME.NAMESPACE=URL, got {entry!r}"
+                )
+            mapping[ref] = url
+
+        def resolver(service_ref):
+            return mapping.get(f"{service_ref.name}.{service_ref.namespace}")
+
+        store.webhook_service_resolver = resolver
+        logger.info("Webhook service resolver: %s", ", ".join(mapping))
 
     server = APIServer(store, port, host="")
     server.start()
