@@ -240,7 +240,8 @@ def version():
 @click.option("--state-file", default="", help="Snapshot file: loaded on start if present, written on shutdown (checkpoint/resume).")
 @click.option("--checkpoint-interval-seconds", default=0.0, show_default=True, help="Also snapshot --state-file every N seconds (crash resilience; 0 = shutdown-only).")
 @click.option("--resolve-webhook-service", "webhook_services", multiple=True, metavar="NAME.NAMESPACE=URL", help="Resolve a ValidatingWebhookConfiguration service reference to a URL (clusters use <name>.<ns>.svc DNS; repeatable).")
-def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services):
+@click.option("--token", default="", help="Require `Authorization: Bearer <token>` on every request except /healthz (static-token authn).")
+def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services, token):
     """Serve the in-memory API store over HTTP (hermetic e2e backend)."""
     import json as jsonlib
 
@@ -271,9 +272,10 @@ def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services):
         store.webhook_service_resolver = resolver
         logger.info("Webhook service resolver: %s", ", ".join(mapping))
 
-    server = APIServer(store, port, host="")
+    server = APIServer(store, port, host="", bearer_token=token or None)
     server.start()
-    logger.info("API server listening on :%d", server.port)
+    logger.info("API server listening on :%d%s", server.port,
+                " (bearer-token authn)" if token else "")
     stop = setup_signal_handler()
     if state_file and checkpoint_interval_seconds > 0:
         store.start_checkpointer(state_file, checkpoint_interval_seconds, stop)

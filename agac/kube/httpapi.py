@@ -69,6 +69,24 @@ class _Handler(BaseHTTPRequestHandler):
     disable_nagle_algorithm = True  # small JSON bodies; latency over batching
     store: APIStore = None  # set by server factory
     watch_idle_seconds = 5.0  # heartbeat/bookmark cadence on idle watches
+    bearer_token = None  # require `Authorization: Bearer <token>` when set
+
+    def _authorized(self) -> bool:
+        """Bearer-token authn (apiserver static-token file shape); /healthz
+        stays open like a real apiserver's healthz with anonymous-auth."""
+        if self.bearer_token is None:
+            return True
+        if self.path.split("?", 1)[0] == "/healthz":
+            return True
+        header = self.headers.get("Authorization", "")
+        if header == f"Bearer {self.bearer_token}":
+            return True
+        self._json(401, {
+            "kind": "Status", "apiVersion": "v1", "status": "Failure",
+            "reason": "Unauthorized", "code": 401,
+            "message": "Unauthorized",
+        })
+        return False
 
     def log_message(self, fmt, *args):  # noqa: A003
         logger.debug(fmt, *args)
@@ -187,6 +205,8 @@ class _Handler(BaseHTTPRequestHandler):
 
     # -- verbs -------------------------------------------------------------
     def do_GET(self):  # noqa: N802
+        if not self._authorized():
+            return
         parts, query = self._route()
         if self._k8s_route(parts, query, "GET"):
             return
@@ -218,6 +238,8 @@ class _Handler(BaseHTTPRequestHandler):
             pass
 
     def do_POST(self):  # noqa: N802
+        if not self._authorized():
+            return
         parts, query = self._route()
         if self._k8s_route(parts, query, "POST"):
             return
@@ -233,6 +255,8 @@ class _Handler(BaseHTTPRequestHandler):
             self._api_error(e)
 
     def do_PUT(self):  # noqa: N802
+        if not self._authorized():
+            return
         parts, query = self._route()
         if self._k8s_route(parts, query, "PUT"):
             return
@@ -254,6 +278,8 @@ class _Handler(BaseHTTPRequestHandler):
             self._api_error(e)
 
     def do_PATCH(self):  # noqa: N802
+        if not self._authorized():
+            return
         parts, query = self._route()
         if self._k8s_route(parts, query, "PATCH"):
             return
@@ -269,6 +295,8 @@ class _Handler(BaseHTTPRequestHandler):
             self._api_error(e)
 
     def do_DELETE(self):  # noqa: N802
+        if not self._authorized():
+            return
         parts, query = self._route()
         if self._k8s_route(parts, query, "DELETE"):
             return
@@ -373,11 +401,12 @@ class APIServer:
     """Owns the HTTP listener for one APIStore."""
 
     def __init__(self, store: APIStore, port: int = 0, host: str = "127.0.0.1",
-                 watch_idle_seconds: float = 5.0):
+                 watch_idle_seconds: float = 5.0, bearer_token=None):
         handler = type(
             "BoundHandler",
             (_Handler,),
-            {"store": store, "watch_idle_seconds": watch_idle_seconds},
+            {"store": store, "watch_idle_seconds": watch_idle_seconds,
+             "bearer_token": bearer_token},
         )
         self.httpd = ThreadingHTTPServer((host, port), handler)
         self.httpd.daemon_threads = True

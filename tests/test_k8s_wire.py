@@ -438,3 +438,95 @@ spec:
         assert live.metadata.labels.get("tier") == "edge"
         # server-populated status survived the apply
         assert live.status.load_balancer.ingress[0].hostname == "lb.example.com"
+
+
+class TestBearerTokenAuthn:
+    """Static-token authn on the hermetic apiserver: K8sKubeClient's
+    Authorization header is actually ENFORCED (not just sent)."""
+
+    def _server(self):
+        from agac.kube.store import APIStore
+
+        srv = APIServer(APIStore(), bearer_token="sekret")
+        srv.start()
+        return srv
+
+    def test_right_token_works_wrong_token_401(self):
+        srv = self._server()
+        try:
+            ok = K8sKubeClient(RestConfig(host=srv.url, token="sekret"))
+            items, _ = ok.list("Service")
+            assert items == []
+
+            from agac.kube.store import APIError
+
+            bad = K8sKubeClient(RestConfig(host=srv.url, token="wrong"))
+            try:
+                bad.list("Service")
+                raise AssertionError("unauthenticated list succeeded")
+            except APIError as e:
+                assert e.code == 401
+            anon = K8sKubeClient(RestConfig(host=srv.url))
+            try:
+                anon.list("Service")
+                raise AssertionError("anonymous list succeeded")
+            except APIError as e:
+                assert e.code == 401
+        finally:
+            srv.shutdown()
+
+    def test_healthz_stays_open(self):
+        import requests
+
+        srv = self._server()
+        try:
+            r = requests.get(f"{srv.url}/healthz", timeout=5)
+            assert r.status_code == 200
+        finally:
+            srv.shutdown()
+
+    def test_full_manager_runs_authenticated(self):
+        import threading
+        import time
+
+        from agac.apis import core as corev1
+        from agac.apis.meta import ObjectMeta
+        from agac.cloudprovider.aws.client import FakeCloudFactory
+        from agac.cloudprovider.fake import FakeAWSBackend
+        from agac.manager import ControllerConfig, Manager
+
+        srv = self._server()
+        backend = FakeAWSBackend()
+        stop = threading.Event()
+        try:
+            client = K8sKubeClient(RestConfig(host=srv.url, token="sekret"))
+            manager = Manager()
+            manager.run(client, ControllerConfig(), FakeCloudFactory(backend),
+                        stop, resync_period=300.0, block=False)
+            assert manager.wait_until_ready()
+            lb = backend.elbv2.create_load_balancer("auth", region="us-east-1")
+            client.create(corev1.Service(
+                metadata=ObjectMeta(
+                    name="auth", namespace="default",
+                    annotations={
+                        "service.beta.kubernetes.io/aws-load-balancer-type": "nlb",
+                        "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-managed": "true",
+                    },
+                ),
+                spec=corev1.ServiceSpec(
+                    type="LoadBalancer",
+                    ports=[corev1.ServicePort(port=80, protocol="TCP")],
+                ),
+                status=corev1.ServiceStatus(
+                    load_balancer=corev1.LoadBalancerStatus(
+                        ingress=[corev1.LoadBalancerIngress(hostname=lb.dns_name)]
+                    )
+                ),
+            ))
+            deadline = time.monotonic() + 15
+            while not backend.ga.list_accelerators()[0]:
+                assert time.monotonic() < deadline
+                time.sleep(0.02)
+        finally:
+            stop.set()
+            srv.shutdown()
