@@ -241,7 +241,9 @@ def version():
 @click.option("--checkpoint-interval-seconds", default=0.0, show_default=True, help="Also snapshot --state-file every N seconds (crash resilience; 0 = shutdown-only).")
 @click.option("--resolve-webhook-service", "webhook_services", multiple=True, metavar="NAME.NAMESPACE=URL", help="Resolve a ValidatingWebhookConfiguration service reference to a URL (clusters use <name>.<ns>.svc DNS; repeatable).")
 @click.option("--token", default="", help="Require `Authorization: Bearer <token>` on every request except /healthz (static-token authn).")
-def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services, token):
+@click.option("--tls-cert-file", default="", help="Serve HTTPS with this certificate (pair with --tls-private-key-file).")
+@click.option("--tls-private-key-file", default="", help="TLS private key.")
+def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services, token, tls_cert_file, tls_private_key_file):
     """Serve the in-memory API store over HTTP (hermetic e2e backend)."""
     import json as jsonlib
 
@@ -272,9 +274,12 @@ def apiserver(port, state_file, checkpoint_interval_seconds, webhook_services, t
         store.webhook_service_resolver = resolver
         logger.info("Webhook service resolver: %s", ", ".join(mapping))
 
-    server = APIServer(store, port, host="", bearer_token=token or None)
+    server = APIServer(store, port, host="", bearer_token=token or None,
+                       tls_cert_file=tls_cert_file,
+                       tls_key_file=tls_private_key_file)
     server.start()
-    logger.info("API server listening on :%d%s", server.port,
+    logger.info("API server listening on :%d%s%s", server.port,
+                " (TLS)" if server.ssl_enabled else "",
                 " (bearer-token authn)" if token else "")
     stop = setup_signal_handler()
     if state_file and checkpoint_interval_seconds > 0:

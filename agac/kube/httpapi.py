@@ -401,7 +401,8 @@ class APIServer:
     """Owns the HTTP listener for one APIStore."""
 
     def __init__(self, store: APIStore, port: int = 0, host: str = "127.0.0.1",
-                 watch_idle_seconds: float = 5.0, bearer_token=None):
+                 watch_idle_seconds: float = 5.0, bearer_token=None,
+                 tls_cert_file: str = "", tls_key_file: str = ""):
         handler = type(
             "BoundHandler",
             (_Handler,),
@@ -410,6 +411,15 @@ class APIServer:
         )
         self.httpd = ThreadingHTTPServer((host, port), handler)
         self.httpd.daemon_threads = True
+        self.ssl_enabled = bool(tls_cert_file and tls_key_file)
+        if self.ssl_enabled:
+            import ssl
+
+            context = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+            context.load_cert_chain(tls_cert_file, tls_key_file)
+            self.httpd.socket = context.wrap_socket(
+                self.httpd.socket, server_side=True
+            )
         self.store = store
 
     @property
@@ -419,7 +429,8 @@ class APIServer:
     @property
     def url(self) -> str:
         host, port = self.httpd.server_address[:2]
-        return f"http://{host}:{port}"
+        scheme = "https" if self.ssl_enabled else "http"
+        return f"{scheme}://{host}:{port}"
 
     def start(self) -> threading.Thread:
         thread = threading.Thread(

@@ -143,6 +143,10 @@ class K8sKubeClient(KubeClient):
         import time as _time
 
         attempt = 0
+        # per-request verify: requests gives REQUESTS_CA_BUNDLE/CURL_CA_BUNDLE
+        # env precedence over session.verify, which would silently override
+        # the kubeconfig's certificate-authority — the explicit kwarg wins
+        kwargs.setdefault("verify", self.session.verify)
         while True:
             r = getattr(self.session, method)(url, **kwargs)
             if r.status_code not in (429, 503) or attempt >= self.max_retries:
@@ -290,6 +294,7 @@ class K8sKubeClient(KubeClient):
             params=params,
             stream=True,
             timeout=(self.timeout, 30.0),
+            verify=self.session.verify,
         )
         _raise_for(r)
         return _K8sWatch(r, kind)
