@@ -150,3 +150,29 @@ def test_vwc_over_the_wire_activates_admission():
         client.create(corev1.Service(metadata=ObjectMeta(name="ok", namespace="default")))
     finally:
         server.shutdown()
+
+
+def test_wildcard_rules_match():
+    from agac.apis.admissionregistration import RuleWithOperations
+    from agac.kube import k8swire
+    from agac.kube.dynamicadmission import _rule_matches
+
+    gvr = k8swire.gvr_for_kind("EndpointGroupBinding")
+    assert _rule_matches(
+        RuleWithOperations(api_groups=["*"], operations=["*"], resources=["*"]),
+        gvr, "DELETE",
+    )
+    assert not _rule_matches(
+        RuleWithOperations(api_groups=["*"], operations=["CREATE"], resources=["*"]),
+        gvr, "DELETE",
+    )
+    assert not _rule_matches(
+        RuleWithOperations(api_groups=["apps"], operations=["*"], resources=["*"]),
+        gvr, "CREATE",
+    )
+    svc = k8swire.gvr_for_kind("Service")
+    assert _rule_matches(
+        RuleWithOperations(api_groups=[""], operations=["UPDATE"],
+                           resources=["services"]),
+        svc, "UPDATE",
+    )
