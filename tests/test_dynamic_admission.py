@@ -176,3 +176,89 @@ def test_wildcard_rules_match():
                            resources=["services"]),
         svc, "UPDATE",
     )
+
+
+def test_webhook_outage_blocks_finalizer_removal_until_policy_relaxed():
+    """The classic cluster incident, reproduced end to end: a Fail-closed
+    webhook goes down; the EGB controller can then no longer remove its
+    finalizer (spec updates are vetoed), so deletes wedge — until the
+    operator flips failurePolicy to Ignore, at which point the drain
+    completes."""
+    import threading
+    import time
+
+    from agac.cloudprovider.aws.client import FakeCloudFactory
+    from agac.cloudprovider.fake import FakeAWSBackend
+    from agac.kube.store import NotFoundError
+    from agac.manager import ControllerConfig, Manager
+
+    store = APIStore()
+    client = InMemoryKubeClient(store)
+    backend = FakeAWSBackend()
+    stop = threading.Event()
+    manager = Manager()
+    manager.run(client, ControllerConfig(), FakeCloudFactory(backend), stop,
+                resync_period=300.0, block=False)
+    try:
+        assert manager.wait_until_ready()
+        # seed an endpoint group + binding while admission is permissive
+        acc = backend.ga.create_accelerator("ext")
+        from agac.cloudprovider.aws import types as t
+
+        listener = backend.ga.create_listener(
+            acc.accelerator_arn, [t.PortRange(80, 80)], "TCP")
+        group = backend.ga.create_endpoint_group(listener.listener_arn, "us-east-1")
+        lb = backend.elbv2.create_load_balancer("wh", region="us-east-1")
+        from agac.apis import core as corev1
+
+        client.create(corev1.Service(
+            metadata=ObjectMeta(name="wh", namespace="default"),
+            spec=corev1.ServiceSpec(type="LoadBalancer"),
+            status=corev1.ServiceStatus(
+                load_balancer=corev1.LoadBalancerStatus(
+                    ingress=[corev1.LoadBalancerIngress(hostname=lb.dns_name)]
+                )
+            ),
+        ))
+        client.create(egb.EndpointGroupBinding(
+            metadata=ObjectMeta(name="whb", namespace="default"),
+            spec=egb.EndpointGroupBindingSpec(
+                endpoint_group_arn=group.endpoint_group_arn,
+                service_ref=egb.ServiceReference(name="wh"),
+            ),
+        ))
+        deadline = time.monotonic() + 15
+        while True:
+            g = backend.ga.describe_endpoint_group(group.endpoint_group_arn)
+            if any(d.endpoint_id == lb.load_balancer_arn
+                   for d in g.endpoint_descriptions):
+                break
+            assert time.monotonic() < deadline
+            time.sleep(0.02)
+
+        # the webhook "goes down": a Fail-closed VWC with no reachable
+        # backend appears (matches UPDATE on endpointgroupbindings)
+        apply_yaml(client, open("config/webhook/manifests.yaml").read())
+
+        client.delete("EndpointGroupBinding", "default", "whb")
+        time.sleep(1.0)  # controller churns: finalizer removal is vetoed
+        live = client.get("EndpointGroupBinding", "default", "whb")
+        assert live.metadata.deletion_timestamp is not None
+        assert live.metadata.finalizers, "finalizer vanished despite Fail-closed webhook"
+
+        # operator escape hatch: relax the policy
+        vwc = client.get("ValidatingWebhookConfiguration", "",
+                         "aws-global-accelerator-controller-validating-webhook")
+        vwc.webhooks[0].failure_policy = "Ignore"
+        client.update(vwc)
+
+        deadline = time.monotonic() + 15
+        while True:
+            try:
+                client.get("EndpointGroupBinding", "default", "whb")
+            except NotFoundError:
+                break
+            assert time.monotonic() < deadline, "drain never completed after relax"
+            time.sleep(0.05)
+    finally:
+        stop.set()
