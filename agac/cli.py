@@ -94,9 +94,10 @@ def resolve_kubeconfig(kubeconfig: str) -> str:
 @click.option("--api", type=click.Choice(["memory", "http", "k8s"]), default="memory", show_default=True, help="Kube API backend: embedded in-memory store, an agac HTTP API server, or a real Kubernetes API server (kubeconfig/in-cluster auth).")
 @click.option("--cloud", type=click.Choice(["auto", "aws", "fake"]), default="auto", show_default=True, help="Cloud backend: auto (aws for http/k8s APIs, fake for memory), aws (boto3, required), fake (in-memory).")
 @click.option("--metrics-port", default=0, help="Serve Prometheus metrics on this port (0 = disabled).")
+@click.option("--health-port", default=0, help="Serve /healthz (liveness) and /readyz (caches synced) on this port (0 = disabled).")
 @click.option("--cloud-resync-minutes", default=0.0, show_default=True, help="Re-enqueue ALL managed objects every N minutes even if unchanged, repairing cloud-side drift (0 = disabled, matching the reference: drift on unchanged objects is never repaired — docs/PARITY.md).")
 @click.option("--leader-elect/--no-leader-elect", default=True, show_default=True)
-def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_port, cloud_resync_minutes, leader_elect):
+def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_port, health_port, cloud_resync_minutes, leader_elect):
     """Start controller."""
     from .controller.endpointgroupbinding import EndpointGroupBindingConfig
     from .controller.globalaccelerator import GlobalAcceleratorConfig
@@ -178,9 +179,27 @@ def controller(workers, cluster_name, kubeconfig, master, api, cloud, metrics_po
     )
     stop = setup_signal_handler()
 
+    current_manager = {}
+    if health_port:
+        from .health import HealthServer
+
+        health = HealthServer(
+            health_port,
+            ready_fn=lambda: (
+                current_manager.get("m") is not None
+                and current_manager["m"].is_ready()
+            ),
+        )
+        health.start()
+        logger.info("Health endpoints on :%d (/healthz, /readyz)", health.port)
+
     def run_manager(stop_leading):
         manager = Manager()
-        manager.run(kube_client, config, cloud_factory, stop_leading, block=True)
+        current_manager["m"] = manager
+        try:
+            manager.run(kube_client, config, cloud_factory, stop_leading, block=True)
+        finally:
+            current_manager.pop("m", None)
 
     if leader_elect:
         elector = LeaderElector(

@@ -134,6 +134,14 @@ class Manager:
             for thread in self.threads:
                 thread.join(timeout=5.0)
 
+    def is_ready(self) -> bool:
+        """Readiness: every controller's informer caches have synced
+        (the /readyz condition for deployment probes)."""
+        if self.informer_factory is None:
+            return False
+        informers = list(self.informer_factory._informers.values())
+        return bool(informers) and all(i.has_synced() for i in informers)
+
     def wait_until_ready(self, timeout: float = 30.0) -> bool:
         """Test helper: wait until every controller's caches have synced and
         workers are pumping."""
