@@ -434,3 +434,54 @@ class TestSchemaEnforcement:
             )
         # first change did NOT commit
         assert env.sims["route53"]._records[zone["Id"]] == {}
+
+
+class TestIAMPolicyCoverage:
+    """config/iam/policy.json (byte-identical to the reference README's
+    policy) must cover every operation the production adapter can issue —
+    a migrating user attaching that policy must never hit AccessDenied."""
+
+    # boto3 method -> IAM action
+    ACTION_FOR = {
+        "describe_load_balancers": "elasticloadbalancing:DescribeLoadBalancers",
+        "create_accelerator": "globalaccelerator:CreateAccelerator",
+        "describe_accelerator": "globalaccelerator:DescribeAccelerator",
+        "list_accelerators": "globalaccelerator:ListAccelerators",
+        "update_accelerator": "globalaccelerator:UpdateAccelerator",
+        "delete_accelerator": "globalaccelerator:DeleteAccelerator",
+        "list_tags_for_resource": "globalaccelerator:ListTagsForResource",
+        "tag_resource": "globalaccelerator:TagResource",
+        "create_listener": "globalaccelerator:CreateListener",
+        "list_listeners": "globalaccelerator:ListListeners",
+        "update_listener": "globalaccelerator:UpdateListener",
+        "delete_listener": "globalaccelerator:DeleteListener",
+        "create_endpoint_group": "globalaccelerator:CreateEndpointGroup",
+        "list_endpoint_groups": "globalaccelerator:ListEndpointGroups",
+        "describe_endpoint_group": "globalaccelerator:DescribeEndpointGroup",
+        "update_endpoint_group": "globalaccelerator:UpdateEndpointGroup",
+        "add_endpoints": "globalaccelerator:AddEndpoints",
+        "remove_endpoints": "globalaccelerator:RemoveEndpoints",
+        "delete_endpoint_group": "globalaccelerator:DeleteEndpointGroup",
+        "list_hosted_zones": "route53:ListHostedZones",
+        "list_hosted_zones_by_name": "route53:ListHostedZonesByName",
+        "list_resource_record_sets": "route53:ListResourceRecordSets",
+        "change_resource_record_sets": "route53:ChangeResourceRecordSets",
+    }
+
+    def test_every_adapter_operation_is_granted(self):
+        import json
+        import re
+
+        policy = json.load(open("config/iam/policy.json"))
+        granted = {a.lower() for a in policy["Statement"][0]["Action"]}
+        source = open("agac/cloudprovider/aws/boto3_adapter.py").read()
+        called = set(re.findall(r"self\.client\.(\w+)", source))
+        assert called, "no adapter operations found"
+        for method in sorted(called):
+            action = self.ACTION_FOR.get(method)
+            assert action is not None, f"unmapped adapter operation {method}"
+            # IAM action names are case-insensitive (the reference policy's
+            # 'ListHostedzonesByName' casing relies on that)
+            assert action.lower() in granted, (
+                f"{method} needs {action}, missing from config/iam/policy.json"
+            )
