@@ -50,7 +50,7 @@ def _ca_bundle_file(ca_bundle_b64: str) -> Optional[str]:
     return path
 
 
-def _rule_matches(rule, gvr, operation: str) -> bool:
+def _rule_matches(rule, gvr, operation: str, subresource=None) -> bool:
     ops = rule.operations or []
     if "*" not in ops and operation not in ops:
         return False
@@ -58,6 +58,12 @@ def _rule_matches(rule, gvr, operation: str) -> bool:
     if "*" not in groups and gvr.group not in groups:
         return False
     resources = rule.resources or []
+    if subresource:
+        # a subresource write matches only "<plural>/<sub>" or "*/<sub>"
+        return (
+            f"{gvr.plural}/{subresource}" in resources
+            or f"*/{subresource}" in resources
+        )
     return "*" in resources or gvr.plural in resources
 
 
@@ -89,7 +95,7 @@ def _call_webhook(webhook, url: str, verify, kind: str, operation: str,
         )
 
 
-def admit(store, kind: str, operation: str, old, new):
+def admit(store, kind: str, operation: str, old, new, subresource=None):
     """Evaluate every registered ValidatingWebhookConfiguration against
     this write.  Raises AdmissionDeniedError to veto."""
     from . import k8swire
@@ -103,7 +109,10 @@ def admit(store, kind: str, operation: str, old, new):
         return
     for config in configs:
         for webhook in config.webhooks:
-            if not any(_rule_matches(r, gvr, operation) for r in webhook.rules):
+            if not any(
+                _rule_matches(r, gvr, operation, subresource)
+                for r in webhook.rules
+            ):
                 continue
             cc = webhook.client_config
             url = cc.url

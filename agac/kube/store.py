@@ -154,7 +154,7 @@ class APIStore:
             self._objects.get("ValidatingWebhookConfiguration")
         )
 
-    def _admit(self, kind: str, operation: str, old, new):
+    def _admit(self, kind: str, operation: str, old, new, subresource=None):
         """Run admission outside the store lock (webhook calls may do
         network I/O); commit-time rv checks still serialize writers.
         Consults both programmatically registered hooks and any
@@ -173,7 +173,8 @@ class APIStore:
                 old_d = metalib.to_dict(old)
             if new_d is None and new is not None:
                 new_d = metalib.to_dict(new)
-            dynamicadmission.admit(self, kind, operation, old_d, new_d)
+            dynamicadmission.admit(self, kind, operation, old_d, new_d,
+                                   subresource=subresource)
 
     # -- helpers -----------------------------------------------------------
     def _bucket(self, kind: str) -> Dict[Tuple[str, str], object]:
@@ -366,8 +367,16 @@ class APIStore:
             return metalib.deep_copy(stored)
 
     def update_status(self, obj):
-        """Status-subresource update: only ``status`` is taken from ``obj``."""
+        """Status-subresource update: only ``status`` is taken from ``obj``.
+        Admission applies only to webhooks whose rules name the status
+        subresource ("<plural>/status"), like a real apiserver."""
         kind = type(obj).kind
+        if self._admission_active():
+            try:
+                current = self.get(kind, obj.metadata.namespace, obj.metadata.name)
+            except NotFoundError:
+                current = None
+            self._admit(kind, "UPDATE", current, obj, subresource="status")
         with self._lock:
             key = (obj.metadata.namespace, obj.metadata.name)
             bucket = self._bucket(kind)

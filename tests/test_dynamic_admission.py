@@ -262,3 +262,46 @@ def test_webhook_outage_blocks_finalizer_removal_until_policy_relaxed():
             time.sleep(0.05)
     finally:
         stop.set()
+
+
+def test_status_subresource_rules(store):
+    """Status writes consult admission only when a rule names
+    <plural>/status (real apiserver subresource matching): the default
+    manifest (spec-only rules) never blocks status updates, and a
+    status-scoped Fail-closed rule does."""
+    store.create(binding(name="sub"))
+    live = store.get("EndpointGroupBinding", "default", "sub")
+    live.status.endpoint_ids = ["arn:lb"]
+    store.update_status(live)  # default manifest: unaffected
+
+    # add a status-scoped webhook with an unreachable backend (Fail)
+    from agac.apis.admissionregistration import (
+        RuleWithOperations,
+        ValidatingWebhook,
+        ValidatingWebhookConfiguration,
+        WebhookClientConfig,
+    )
+
+    store.create(ValidatingWebhookConfiguration(
+        metadata=ObjectMeta(name="status-guard"),
+        webhooks=[ValidatingWebhook(
+            name="status.example.com",
+            client_config=WebhookClientConfig(url="https://127.0.0.1:1/x"),
+            rules=[RuleWithOperations(
+                api_groups=["operator.h3poteto.dev"],
+                operations=["UPDATE"],
+                resources=["endpointgroupbindings/status"],
+            )],
+            failure_policy="Fail",
+            timeout_seconds=1,
+        )],
+    ))
+    live = store.get("EndpointGroupBinding", "default", "sub")
+    live.status.endpoint_ids = ["arn:lb2"]
+    with pytest.raises(AdmissionDeniedError):
+        store.update_status(live)
+    # spec updates don't match the status-scoped rule (only the default
+    # manifest's spec rule applies, and the weight change is allowed)
+    live = store.get("EndpointGroupBinding", "default", "sub")
+    live.spec.weight = 44
+    store.update(live)
