@@ -502,17 +502,25 @@ class GlobalAcceleratorMixin:
             self._delete_accelerator(accelerator.accelerator_arn)
 
     def _list_related(self, arn: str):
+        """Deliberate fix (docs/PARITY.md §5c): the reference's
+        listRelatedGlobalAccelerator treats ANY error as "already deleted"
+        (global_accelerator.go:274-288 returns nil on err), so a single
+        throttled DescribeAccelerator during teardown makes
+        CleanupGlobalAccelerator report success and the accelerator leaks
+        FOREVER (nothing re-enqueues a forgotten cleanup).  Only the typed
+        not-found errors mean "gone" — anything else propagates so the
+        engine's rate-limited retry runs the cleanup again."""
         try:
             accelerator = self._get_accelerator(arn)
-        except Exception:
+        except awserr.AcceleratorNotFoundException:
             return None, None, None
         try:
             listener = self.get_listener(accelerator.accelerator_arn)
-        except Exception:
+        except awserr.ListenerNotFoundException:
             return accelerator, None, None
         try:
             endpoint_group = self.get_endpoint_group(listener.listener_arn)
-        except Exception:
+        except awserr.EndpointGroupNotFoundException:
             return accelerator, listener, None
         return accelerator, listener, endpoint_group
 

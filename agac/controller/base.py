@@ -137,11 +137,17 @@ def spawn_cloud_resync(
     return thread
 
 
-def make_queue_rate_limiter(qps: float, burst: int):
+def make_queue_rate_limiter(qps: float, burst: int,
+                            item_base_delay: float = 0.005,
+                            item_max_delay: float = 1000.0):
     """Controller queue limiter: per-item exponential backoff + overall
     token bucket (client-go DefaultControllerRateLimiter shape) with
     configurable qps/burst — the reference hardcodes 10 qps / 100 burst,
-    which caps sustained reconcile throughput at 10 objects/s."""
+    which caps sustained reconcile throughput at 10 objects/s.  The
+    per-item backoff bounds are the client-go
+    NewItemExponentialFailureRateLimiter knobs (defaults 5ms..1000s; after
+    a sustained AWS throttle storm the 1000s cap bounds how long a single
+    object can wait before its next retry)."""
     from ..kube.workqueue import (
         BucketRateLimiter,
         ItemExponentialFailureRateLimiter,
@@ -149,5 +155,6 @@ def make_queue_rate_limiter(qps: float, burst: int):
     )
 
     return MaxOfRateLimiter(
-        ItemExponentialFailureRateLimiter(), BucketRateLimiter(qps=qps, burst=burst)
+        ItemExponentialFailureRateLimiter(item_base_delay, item_max_delay),
+        BucketRateLimiter(qps=qps, burst=burst),
     )

@@ -48,6 +48,9 @@ class EndpointGroupBindingConfig:
     workers: int = 1
     queue_qps: float = 10.0
     queue_burst: int = 100
+    # per-item failure-backoff bounds (client-go defaults)
+    queue_item_base_delay: float = 0.005
+    queue_item_max_delay: float = 1000.0
     # opt-in drift repair (see docs/PARITY.md §resync); 0 = parity
     cloud_resync_period: float = 0.0
 
@@ -63,7 +66,7 @@ class EndpointGroupBindingController:
         self.cloud_factory = cloud_factory
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
         self.workqueue = RateLimitingQueue(
-            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
+            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst, config.queue_item_base_delay, config.queue_item_max_delay),
             name="EndpointGroupBinding",
         )
 

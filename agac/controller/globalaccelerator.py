@@ -48,6 +48,9 @@ class GlobalAcceleratorConfig:
     # queue token-bucket rate (client-go default 10/100); raise for scale
     queue_qps: float = 10.0
     queue_burst: int = 100
+    # per-item failure-backoff bounds (client-go defaults)
+    queue_item_base_delay: float = 0.005
+    queue_item_max_delay: float = 1000.0
     # opt-in drift repair: re-enqueue unchanged managed objects every N
     # seconds (0 = reference-parity behavior: cloud drift on an unchanged
     # object is never repaired — see docs/PARITY.md §resync)
@@ -68,11 +71,11 @@ class GlobalAcceleratorController:
         self._hints_lock = threading.Lock()
         self.recorder = EventRecorder(kube_client, CONTROLLER_AGENT_NAME)
         self.service_queue = RateLimitingQueue(
-            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
+            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst, config.queue_item_base_delay, config.queue_item_max_delay),
             name=CONTROLLER_AGENT_NAME + "-service",
         )
         self.ingress_queue = RateLimitingQueue(
-            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst),
+            rate_limiter=make_queue_rate_limiter(config.queue_qps, config.queue_burst, config.queue_item_base_delay, config.queue_item_max_delay),
             name=CONTROLLER_AGENT_NAME + "-ingress",
         )
 

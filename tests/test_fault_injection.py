@@ -307,3 +307,67 @@ class TestRoute53PartialPair:
         )])
         with pytest.raises(awserr.InvalidChangeBatch):
             cloud._create_metadata_record_set(stored_zone, "b.own.example.com", owner)
+
+
+class TestCleanupLeakUnderThrottle:
+    """PARITY §5c: one throttled DescribeAccelerator during teardown made
+    cleanup report success and leak the accelerator forever (the reference
+    shares this: listRelatedGlobalAccelerator treats ANY error as
+    "already deleted").  Now transient errors propagate so the cleanup is
+    retried to completion."""
+
+    def test_throttled_discovery_does_not_fake_success(self):
+        backend = FakeAWSBackend()
+        cloud = FakeCloudFactory(backend)(REGION)
+        lb = backend.elbv2.create_load_balancer("leak", region=REGION)
+        svc = managed_service("leak", lb)
+        arn, created, _ = cloud.ensure_global_accelerator_for_service(
+            svc, corev1.LoadBalancerIngress(hostname=lb.dns_name), "c",
+            "leak", REGION,
+        )
+        assert created
+
+        def hook(service, op):
+            if op == "describe_accelerator":
+                raise Throttle("injected during teardown discovery")
+
+        backend.set_fault_hook(hook)
+        with pytest.raises(awserr.AWSAPIError):
+            cloud.cleanup_global_accelerator(arn)  # must NOT fake success
+        # accelerator still there (nothing silently forgotten)
+        backend.set_fault_hook(None)
+        assert len(backend.ga.list_accelerators()[0]) == 1
+        # fault clears; the retried cleanup completes
+        cloud.cleanup_global_accelerator(arn)
+        assert backend.ga.list_accelerators()[0] == []
+
+    def test_controller_retries_cleanup_to_completion(self):
+        """End to end: unmanage during a throttle storm on describe; the
+        rate-limited requeue must eventually delete the accelerator
+        instead of leaking it."""
+        backend = FakeAWSBackend()
+        client, stop = start_stack(backend)
+        try:
+            lb = backend.elbv2.create_load_balancer("ctl", region=REGION)
+            client.create(managed_service("ctl", lb))
+            wait_for(lambda: len(backend.ga.list_accelerators()[0]) == 1, "create")
+
+            state = {"fail": True}
+
+            def hook(service, op):
+                if state["fail"] and op == "describe_accelerator":
+                    raise Throttle("storm")
+
+            backend.set_fault_hook(hook)
+            svc = client.get("Service", "default", "ctl")
+            del svc.metadata.annotations[
+                "aws-global-accelerator-controller.h3poteto.dev/global-accelerator-managed"
+            ]
+            client.update(svc)
+            time.sleep(0.5)  # cleanup attempts keep failing (and retrying)
+            assert len(backend.ga.list_accelerators()[0]) == 1
+            state["fail"] = False
+            wait_for(lambda: len(backend.ga.list_accelerators()[0]) == 0,
+                     "retried cleanup after storm clears")
+        finally:
+            stop.set()
