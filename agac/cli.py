@@ -330,6 +330,9 @@ def get_cmd(kind, namespace, master):
         "endpointgroupbindings": "EndpointGroupBinding", "egb": "EndpointGroupBinding",
         "lease": "Lease", "leases": "Lease",
         "event": "Event", "events": "Event",
+        "validatingwebhookconfiguration": "ValidatingWebhookConfiguration",
+        "validatingwebhookconfigurations": "ValidatingWebhookConfiguration",
+        "vwc": "ValidatingWebhookConfiguration",
     }
     resolved = aliases.get(kind.lower())
     if resolved is None:
@@ -345,6 +348,10 @@ def get_cmd(kind, namespace, master):
             detail = f"endpoints={len(obj.status.endpoint_ids)} gen={obj.metadata.generation} observed={obj.status.observed_generation}"
         elif resolved == "Lease":
             detail = f"holder={obj.spec.holder_identity}"
+        elif resolved == "ValidatingWebhookConfiguration":
+            detail = ", ".join(
+                f"{w.name}({w.failure_policy})" for w in obj.webhooks
+            )
         elif resolved == "Event":
             detail = f"{obj.reason} x{obj.count}"
         click.echo(
