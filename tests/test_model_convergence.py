@@ -591,3 +591,159 @@ def test_concentrated_storm_on_one_op_family_converges(family):
     offline; one representative seed pinned here."""
     run_faulty_walk(400007, n_services=6, n_ops=50, fault_rate=0.7,
                     target_ops=OP_FAMILIES[family])
+
+
+def run_drift_walk(seed: int, n_services: int = 6, n_ops: int = 60):
+    """Walks with an extra adversary: ops that corrupt the CLOUD directly
+    (listener ports, tags) behind the controllers' backs.  With
+    cloud-resync enabled, the audit must converge to the model anyway —
+    the sustained version of tests/test_cloud_resync.py."""
+    rng = random.Random(seed)
+    backend = FakeAWSBackend()
+    client = InMemoryKubeClient()
+    stop = threading.Event()
+    manager = Manager()
+    from agac.controller.endpointgroupbinding import EndpointGroupBindingConfig
+    from agac.controller.globalaccelerator import GlobalAcceleratorConfig
+    from agac.controller.route53 import Route53Config
+
+    config = ControllerConfig(
+        global_accelerator=GlobalAcceleratorConfig(cloud_resync_period=0.2),
+        route53=Route53Config(cloud_resync_period=0.2),
+        endpoint_group_binding=EndpointGroupBindingConfig(cloud_resync_period=0.2),
+    )
+    factory = FakeCloudFactory(backend, ga_missing_retry=0.1)
+    manager.run(client, config, factory, stop, resync_period=300.0, block=False)
+    assert manager.wait_until_ready()
+    model = Model()
+    backend.route53.create_hosted_zone("walk.example.com")
+    lbs = {}
+
+    def k8s_service(name):
+        m = model.services[name]
+        annotations = {LB_TYPE: "nlb"}
+        if m["managed"]:
+            annotations[MANAGED] = "true"
+        return corev1.Service(
+            metadata=ObjectMeta(name=name, namespace="default",
+                                annotations=annotations),
+            spec=corev1.ServiceSpec(
+                type="LoadBalancer",
+                ports=[corev1.ServicePort(port=p, protocol="TCP")
+                       for p in m["ports"]],
+            ),
+            status=corev1.ServiceStatus(
+                load_balancer=corev1.LoadBalancerStatus(
+                    ingress=[corev1.LoadBalancerIngress(
+                        hostname=lbs[name].dns_name)]
+                )
+            ),
+        )
+
+    def push(name):
+        desired = k8s_service(name)
+        for _ in range(20):
+            try:
+                live = client.get("Service", "default", name)
+            except NotFoundError:
+                client.create(desired)
+                return
+            live.metadata.annotations = desired.metadata.annotations
+            live.spec.ports = desired.spec.ports
+            try:
+                client.update(live)
+                return
+            except Exception:
+                continue
+
+    def corrupt_cloud():
+        """Mutate a random accelerator's listener or tags directly."""
+        accs, _ = backend.ga.list_accelerators()
+        if not accs:
+            return
+        acc = rng.choice(accs)
+        listeners, _ = backend.ga.list_listeners(acc.accelerator_arn)
+        if listeners and rng.random() < 0.7:
+            from agac.cloudprovider.aws import types as t
+
+            backend.ga.update_listener(
+                listeners[0].listener_arn,
+                port_ranges=[t.PortRange(65000 + rng.randrange(100), 65000)],
+            )
+        else:
+            from agac.cloudprovider.aws import types as t
+
+            backend.ga.tag_resource(acc.accelerator_arn, [
+                t.Tag("aws-global-accelerator-target-hostname", "corrupted")
+            ])
+
+    try:
+        for _ in range(n_ops):
+            op = rng.choice(["create", "manage", "unmanage", "ports",
+                             "delete", "corrupt", "corrupt"])
+            if op == "corrupt":
+                corrupt_cloud()
+                continue
+            name = f"walk-{rng.randrange(n_services)}"
+            if name not in model.services:
+                if op == "delete":
+                    continue
+                if name not in lbs:
+                    lbs[name] = backend.elbv2.create_load_balancer(
+                        name, region=REGION)
+                model.services[name] = {
+                    "managed": False, "ports": [80], "hostname": None}
+                push(name)
+                continue
+            m = model.services[name]
+            if op == "create":
+                continue
+            if op == "manage":
+                m["managed"] = True
+            elif op == "unmanage":
+                m["managed"] = False
+            elif op == "ports":
+                m["ports"] = sorted(rng.sample(range(1000, 1010),
+                                               rng.randint(1, 3)))
+            elif op == "delete":
+                client.delete("Service", "default", name)
+                del model.services[name]
+                continue
+            push(name)
+
+        def audit():
+            managed = model.managed()
+            accs, _ = backend.ga.list_accelerators()
+            owned = {}
+            for a in accs:
+                tags = {t.key: t.value for t in
+                        backend.ga.list_tags_for_resource(a.accelerator_arn)}
+                owner = tags.get(OWNER_TAG)
+                if owner is None:
+                    return False
+                owned.setdefault(owner, []).append((a, tags))
+            if set(owned) != {f"service/default/{n}" for n in managed}:
+                return False
+            for name, m in managed.items():
+                entries = owned[f"service/default/{name}"]
+                if len(entries) != 1:
+                    return False
+                acc, tags = entries[0]
+                if tags.get(HOSTNAME_TAG) != lbs[name].dns_name:
+                    return False  # corrupted tag must be repaired
+                listeners, _ = backend.ga.list_listeners(acc.accelerator_arn)
+                if len(listeners) != 1:
+                    return False
+                if sorted(p.from_port for p in listeners[0].port_ranges) != \
+                        sorted(m["ports"]):
+                    return False  # corrupted ports must be repaired
+            return True
+
+        settle(audit, timeout=30.0, what=f"drift walk seed {seed}")
+    finally:
+        stop.set()
+
+
+@pytest.mark.parametrize("seed", [55, 808, 31415])
+def test_drift_walk_self_heals_with_cloud_resync(seed):
+    run_drift_walk(seed)
