@@ -62,6 +62,11 @@ if _AVAILABLE:
         "Rate-limited re-adds (client-go workqueue_retries_total)",
         ["queue"],
     )
+    WEBHOOK_REVIEWS = Counter(
+        "agac_webhook_reviews_total",
+        "AdmissionReview verdicts served by the webhook process",
+        ["operation", "verdict"],
+    )
 
 
 # prometheus_client's .labels() re-validates and re-hashes on every call;
@@ -114,6 +119,11 @@ def observe_work_duration(queue_name: str, seconds: float):
 def count_queue_retry(queue_name: str):
     if _AVAILABLE and queue_name:
         _child(WORKQUEUE_RETRIES, queue_name).inc()
+
+
+def observe_webhook_review(operation: str, verdict: str):
+    if _AVAILABLE:
+        _child(WEBHOOK_REVIEWS, operation or "UNKNOWN", verdict).inc()
 
 
 def start_metrics_server(port: int):

@@ -48,6 +48,13 @@ class WebhookHandler(BaseHTTPRequestHandler):
             self._write(400, err.encode(), "text/plain")
             return
         response = validator.validate(review)
+        from .. import metrics
+
+        allowed = ((response.get("response") or {}).get("allowed", False))
+        metrics.observe_webhook_review(
+            (review.get("request") or {}).get("operation", ""),
+            "allowed" if allowed else "denied",
+        )
         self._write(200, json.dumps(response).encode())
 
     def _parse_request(self) -> Tuple[Optional[dict], Optional[str]]:
