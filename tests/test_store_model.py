@@ -123,7 +123,7 @@ def apply_op(store, model, op):
 
 def test_all_deletion_schedules_match_the_model():
     n = 0
-    for length in range(1, 6):
+    for length in range(1, 7):
         for schedule in itertools.product(OPS, repeat=length):
             store = APIStore()
             model = Model()
@@ -133,7 +133,7 @@ def test_all_deletion_schedules_match_the_model():
                     schedule, op, visible(store), model_visible(model)
                 )
             n += 1
-    assert n == sum(5**k for k in range(1, 6))  # 3,905 schedules
+    assert n == sum(5**k for k in range(1, 7))  # 19,530 schedules
 
 
 def test_deletion_is_sticky():

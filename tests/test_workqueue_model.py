@@ -109,13 +109,13 @@ def run_schedule(schedule):
     return trace_real
 
 
-def test_all_schedules_up_to_length_7_match_the_model():
+def test_all_schedules_up_to_length_8_match_the_model():
     n = 0
-    for length in range(1, 8):
+    for length in range(1, 9):
         for schedule in itertools.product(OPS, repeat=length):
             run_schedule(schedule)
             n += 1
-    assert n == sum(4**k for k in range(1, 8))  # 21,844 schedules
+    assert n == sum(4**k for k in range(1, 9))  # 87,380 schedules
 
 
 def test_redelivery_after_done_exactly_once():
