@@ -131,7 +131,19 @@ class TestFullLifecycle:
             client.create(svc)
 
             # GA triple appears (local_e2e waitUntilGlobalAccelerator analogue)
-            assert wait_until(lambda: len(backend.ga.list_accelerators()[0]) == 1)
+            def full_triple():
+                accs, _ = backend.ga.list_accelerators()
+                if len(accs) != 1:
+                    return False
+                ls, _ = backend.ga.list_listeners(accs[0].accelerator_arn)
+                if len(ls) != 1:
+                    return False
+                gs, _ = backend.ga.list_endpoint_groups(ls[0].listener_arn)
+                return len(gs) == 1
+
+            # wait for the whole triple: the accelerator appears before its
+            # listener/endpoint group mid-create
+            assert wait_until(full_triple)
             acc = backend.ga.list_accelerators()[0][0]
             listeners, _ = backend.ga.list_listeners(acc.accelerator_arn)
             assert [p.from_port for p in listeners[0].port_ranges] == [80]

@@ -83,12 +83,20 @@ def test_full_scenario_over_production_client(env):
     lb = env.backend.elbv2.create_load_balancer("k8se2e", region=REGION)
     env.client.create(managed_service("k8se2e", lb, "app.k8s.example.com"))
 
-    wait_for(lambda: len(env.backend.ga.list_accelerators()[0]) == 1,
-             "accelerator triple")
+    def full_triple():
+        accs, _ = env.backend.ga.list_accelerators()
+        if len(accs) != 1:
+            return False
+        listeners, _ = env.backend.ga.list_listeners(accs[0].accelerator_arn)
+        if len(listeners) != 1:
+            return False
+        groups, _ = env.backend.ga.list_endpoint_groups(listeners[0].listener_arn)
+        return len(groups) == 1
+
+    wait_for(full_triple, "accelerator triple")
     acc = env.backend.ga.list_accelerators()[0][0]
     listeners, _ = env.backend.ga.list_listeners(acc.accelerator_arn)
     groups, _ = env.backend.ga.list_endpoint_groups(listeners[0].listener_arn)
-    assert len(listeners) == 1 and len(groups) == 1
 
     def record_types():
         recs, _ = env.backend.route53.list_resource_record_sets(zone.id)
