@@ -182,7 +182,15 @@ class TestGlobalAcceleratorIngressPath:
             ),
         )
         client.create(ingress)
-        assert wait_until(lambda: len(list_accelerators(backend)) == 1)
+
+        def accelerator_with_listener():
+            accs = list_accelerators(backend)
+            if len(accs) != 1:
+                return False
+            ls, _ = backend.ga.list_listeners(accs[0].accelerator_arn)
+            return len(ls) == 1  # the triple is created in steps
+
+        assert wait_until(accelerator_with_listener)
         acc = list_accelerators(backend)[0]
         assert acc.name == "ingress-default-ing"
         listeners, _ = backend.ga.list_listeners(acc.accelerator_arn)
