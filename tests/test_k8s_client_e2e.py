@@ -137,10 +137,17 @@ def test_full_scenario_over_production_client(env):
 
     wait_for(attached, "EGB attach")
 
-    # status round-tripped through the wire (observedGeneration gate)
-    binding = env.client.get("EndpointGroupBinding", "default", "k8sbind")
-    assert binding.status.endpoint_ids == [lb2.load_balancer_arn]
-    assert binding.status.observed_generation == binding.metadata.generation
+    # status round-trips through the wire AFTER the cloud attach (the
+    # controller writes endpoint_ids + observedGeneration as a second
+    # step) — wait for it rather than racing it
+    def status_round_tripped():
+        b = env.client.get("EndpointGroupBinding", "default", "k8sbind")
+        return (
+            b.status.endpoint_ids == [lb2.load_balancer_arn]
+            and b.status.observed_generation == b.metadata.generation
+        )
+
+    wait_for(status_round_tripped, "EGB status round-trip")
 
     env.client.delete("EndpointGroupBinding", "default", "k8sbind")
 
