@@ -154,7 +154,10 @@ def test_route53_drift_repaired():
                     return r.alias_target.dns_name
             return None
 
-        wait_for(lambda: alias_dns() is not None, "initial route53 converge")
+        # 90s: covers the reference's 60s GA-missing requeue if the
+        # route53 reconcile races ahead of the accelerator creation
+        wait_for(lambda: alias_dns() is not None, "initial route53 converge",
+                 timeout=90.0)
         good = alias_dns()
 
         # drift the alias record out from under the controller

@@ -102,10 +102,13 @@ def test_full_scenario_over_production_client(env):
         recs, _ = env.backend.route53.list_resource_record_sets(zone.id)
         return {(r.name, r.type) for r in recs}
 
+    # 90s: if the route53 reconcile raced ahead of the accelerator it
+    # requeues at the reference's 60s GA-missing interval
     wait_for(
         lambda: {("app.k8s.example.com.", "A"),
                  ("app.k8s.example.com.", "TXT")} <= record_types(),
         "route53 pair",
+        timeout=90.0,
     )
 
     # EGB attach + drain through the CRD path
